@@ -1,0 +1,181 @@
+"""Pod scheduler for the lws_amd cluster substrate.
+
+Implements the scheduling features the LWS/DS orchestration layer depends
+on (the reference delegates these to kube-scheduler + volcano):
+
+ - nodeSelector matching (exclusive-placement worker pinning,
+   reference pod_controller.go:297-331)
+ - amd.com/gpu resource fit against node capacity
+ - required podAffinity / podAntiAffinity with topologyKey, including the
+   k8s special case that a required podAffinity whose selector matches no
+   existing pod admits a pod that matches its own selector (this is what
+   lets the first pod of an exclusive-topology group schedule)
+ - gang scheduling via PodGroup minMember: pods annotated with a
+   pod-group name are bound all-or-nothing
+   (reference pkg/schedulerprovider/volcano_provider.go)
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+from ..api.core import Pod
+from ..api.meta import selector_matches
+from .controller import Controller, Manager
+from .node import Node, node_gpu_capacity, pod_gpu_request
+from .store import ConflictError, NotFoundError, Store
+
+POD_GROUP_ANNOTATION = "scheduling.k8s.io/group-name"
+
+
+class Scheduler:
+    def __init__(self, manager: Manager, nodes: list[Node]) -> None:
+        self.store: Store = manager.store
+        self.nodes = {n.metadata.name: n for n in nodes}
+        self.ctrl = Controller("scheduler", self.reconcile)
+        manager.add_controller(self.ctrl)
+        manager.watch("Pod", self.ctrl, self._map_pod)
+        manager.watch("PodGroup", self.ctrl,
+                      lambda ev, pg: [(pg.metadata.namespace, "_gang_")])
+
+    def _map_pod(self, event: str, pod) -> list[tuple[str, str]]:
+        # any pod event can unblock pending pods (capacity freed / affinity
+        # satisfiable) — reschedule the namespace's pending set
+        return [(pod.metadata.namespace, "_pending_")]
+
+    # ------------------------------------------------------------------
+    def reconcile(self, namespace: str, _key: str) -> Optional[float]:
+        pending = [p for p in self.store.list("Pod", namespace)
+                   if not p.node_name and p.metadata.deletion_timestamp is None]
+        if not pending:
+            return None
+        scheduled = [p for p in self.store.list("Pod", namespace) if p.node_name]
+
+        # partition into gang groups and singletons
+        gangs: dict[str, list[Pod]] = {}
+        singles: list[Pod] = []
+        for p in pending:
+            g = (p.metadata.annotations or {}).get(POD_GROUP_ANNOTATION, "")
+            if g:
+                gangs.setdefault(g, []).append(p)
+            else:
+                singles.append(p)
+
+        progress = False
+        for p in sorted(singles, key=lambda x: x.metadata.name):
+            node = self._find_node(p, scheduled)
+            if node is not None:
+                self._bind(p, node)
+                scheduled.append(p)
+                progress = True
+
+        for gname, pods in sorted(gangs.items()):
+            pg = self.store.try_get("PodGroup", namespace, gname)
+            min_member = pg.spec.min_member if pg else len(pods)
+            bound_members = [p for p in scheduled
+                             if (p.metadata.annotations or {}).get(
+                                 POD_GROUP_ANNOTATION) == gname]
+            if len(pods) + len(bound_members) < min_member:
+                continue  # gang incomplete — wait for remaining pods
+            # trial assignment for the whole gang
+            trial_scheduled = list(scheduled)
+            assignment: list[tuple[Pod, Node]] = []
+            ok = True
+            for p in sorted(pods, key=lambda x: x.metadata.name):
+                node = self._find_node(p, trial_scheduled)
+                if node is None:
+                    ok = False
+                    break
+                p.node_name = node.metadata.name  # trial-local
+                assignment.append((p, node))
+                trial_scheduled.append(p)
+            if ok:
+                for p, node in assignment:
+                    self._bind(p, node)
+                    scheduled.append(p)
+                progress = True
+        # retry while pods remain pending (bounded churn; cheap scan)
+        still = [p for p in self.store.list("Pod", namespace)
+                 if not p.node_name and p.metadata.deletion_timestamp is None]
+        if still and not progress:
+            return 0.1
+        if still:
+            return 0.01
+        return None
+
+    # ------------------------------------------------------------------
+    def _bind(self, pod: Pod, node: Node) -> None:
+        for _ in range(10):
+            cur = self.store.try_get("Pod", pod.metadata.namespace,
+                                     pod.metadata.name)
+            if cur is None or cur.node_name:
+                return
+            cur.node_name = node.metadata.name
+            cur.status.node_name = node.metadata.name
+            try:
+                self.store.update(cur)
+                return
+            except ConflictError:
+                continue
+
+    def _find_node(self, pod: Pod, scheduled: list[Pod]) -> Optional[Node]:
+        for name in sorted(self.nodes):
+            node = self.nodes[name]
+            if self._fits(pod, node, scheduled):
+                return node
+        return None
+
+    def _fits(self, pod: Pod, node: Node, scheduled: list[Pod]) -> bool:
+        # nodeSelector
+        for k, v in (pod.spec.node_selector or {}).items():
+            if node.metadata.labels.get(k) != v:
+                return False
+        # GPU resource fit
+        req = pod_gpu_request(pod)
+        if req > 0:
+            used = sum(pod_gpu_request(p) for p in scheduled
+                       if p.node_name == node.metadata.name
+                       and p.metadata.deletion_timestamp is None)
+            if used + req > node_gpu_capacity(node):
+                return False
+        # affinity
+        aff = pod.spec.affinity
+        if aff is not None:
+            if aff.pod_affinity is not None:
+                for term in aff.pod_affinity.required_during_scheduling_ignored_during_execution:
+                    if not self._affinity_term_ok(pod, node, term, scheduled,
+                                                  anti=False):
+                        return False
+            if aff.pod_anti_affinity is not None:
+                for term in aff.pod_anti_affinity.required_during_scheduling_ignored_during_execution:
+                    if not self._affinity_term_ok(pod, node, term, scheduled,
+                                                  anti=True):
+                        return False
+        return True
+
+    def _affinity_term_ok(self, pod: Pod, node: Node, term, scheduled,
+                          anti: bool) -> bool:
+        topo_key = term.topology_key
+        node_topo = node.metadata.labels.get(topo_key)
+        if node_topo is None:
+            return not anti and False  # node without the topology label
+        matching = [p for p in scheduled
+                    if p.metadata.namespace == pod.metadata.namespace
+                    and p.metadata.deletion_timestamp is None
+                    and selector_matches(term.label_selector,
+                                         p.metadata.labels or {})]
+        same_domain = [p for p in matching
+                       if self._node_topo(p.node_name, topo_key) == node_topo]
+        if anti:
+            return not same_domain
+        if not matching:
+            # k8s special case: no pod matches anywhere — admit if the
+            # incoming pod matches its own affinity selector
+            return selector_matches(term.label_selector,
+                                    pod.metadata.labels or {})
+        return bool(same_domain)
+
+    def _node_topo(self, node_name: str, key: str) -> Optional[str]:
+        node = self.nodes.get(node_name)
+        if node is None:
+            return None
+        return node.metadata.labels.get(key)

@@ -1,0 +1,359 @@
+"""In-process versioned object store — the lws_amd control-plane substrate.
+
+The reference (kubernetes-sigs/lws) delegates object storage, admission,
+watches, optimistic concurrency and owner-reference garbage collection to
+kube-apiserver + etcd + controller-runtime informer caches.
+lws_amd is standalone, so this module provides those semantics in-process:
+
+ - create/get/list/update/update_status/apply/delete with resourceVersion
+   optimistic concurrency (Conflict on stale writes, like client-go)
+ - generation bump on spec change only (status writes don't bump)
+ - admission chain: registered mutators ("defaulting webhooks") then
+   validators run inside create/update
+ - watch event fan-out to registered handlers (informer analogue)
+ - finalizers + foreground/background cascading deletion via ownerReferences
+   (needed by the restart-policy engine's foreground leader delete,
+   reference pkg/controllers/pod_controller.go:258-263)
+"""
+from __future__ import annotations
+
+import threading
+import time
+from typing import Any, Callable, Optional
+
+from ..api import serde
+from ..api.meta import ObjectMeta
+
+FOREGROUND_FINALIZER = "foregroundDeletion"
+
+ADDED = "ADDED"
+MODIFIED = "MODIFIED"
+DELETED = "DELETED"
+
+
+class ApiError(Exception):
+    def __init__(self, reason: str, message: str = ""):
+        super().__init__(f"{reason}: {message}")
+        self.reason = reason
+        self.message = message
+
+
+class NotFoundError(ApiError):
+    def __init__(self, message: str = ""):
+        super().__init__("NotFound", message)
+
+
+class ConflictError(ApiError):
+    def __init__(self, message: str = ""):
+        super().__init__("Conflict", message)
+
+
+class AlreadyExistsError(ApiError):
+    def __init__(self, message: str = ""):
+        super().__init__("AlreadyExists", message)
+
+
+class InvalidError(ApiError):
+    def __init__(self, message: str = ""):
+        super().__init__("Invalid", message)
+
+
+def obj_kind(obj: Any) -> str:
+    return getattr(obj, "kind", type(obj).__name__)
+
+
+def obj_key(obj: Any) -> tuple[str, str, str]:
+    return (obj_kind(obj), obj.metadata.namespace, obj.metadata.name)
+
+
+class Store:
+    def __init__(self) -> None:
+        self._lock = threading.RLock()
+        self._objects: dict[tuple[str, str, str], Any] = {}
+        self._rv = 0
+        self._uid = 0
+        # kind -> list of fn(event_type, obj)
+        self._handlers: dict[str, list[Callable[[str, Any], None]]] = {}
+        self._all_handlers: list[Callable[[str, Any], None]] = []
+        # kind -> list of mutator fn(obj)        (defaulting webhooks)
+        self._mutators: dict[str, list[Callable[[Any], None]]] = {}
+        # kind -> list of validator fn(obj, old) (validating webhooks)
+        self._validators: dict[str, list[Callable[[Any, Optional[Any]], None]]] = {}
+
+    # -- admission / watch registration ------------------------------------
+    def add_mutator(self, kind: str, fn: Callable[[Any], None]) -> None:
+        self._mutators.setdefault(kind, []).append(fn)
+
+    def add_validator(self, kind: str, fn: Callable[[Any, Optional[Any]], None]) -> None:
+        self._validators.setdefault(kind, []).append(fn)
+
+    def add_handler(self, kind: Optional[str], fn: Callable[[str, Any], None]) -> None:
+        """Register a watch handler; kind=None receives all events."""
+        if kind is None:
+            self._all_handlers.append(fn)
+        else:
+            self._handlers.setdefault(kind, []).append(fn)
+
+    def _dispatch(self, events: list[tuple[str, Any]]) -> None:
+        for ev, obj in events:
+            for fn in self._handlers.get(obj_kind(obj), []):
+                fn(ev, serde.deep_copy(obj))
+            for fn in self._all_handlers:
+                fn(ev, serde.deep_copy(obj))
+
+    # -- core verbs --------------------------------------------------------
+    def create(self, obj: Any) -> Any:
+        obj = serde.deep_copy(obj)
+        kind = obj_kind(obj)
+        for fn in self._mutators.get(kind, []):
+            fn(obj)
+        for fn in self._validators.get(kind, []):
+            fn(obj, None)
+        events: list[tuple[str, Any]] = []
+        with self._lock:
+            key = obj_key(obj)
+            if not obj.metadata.name:
+                raise InvalidError("metadata.name required")
+            if key in self._objects:
+                raise AlreadyExistsError(f"{key}")
+            self._uid += 1
+            self._rv += 1
+            obj.metadata.uid = f"uid-{self._uid}"
+            obj.metadata.resource_version = str(self._rv)
+            obj.metadata.generation = 1
+            obj.metadata.creation_timestamp = time.time()
+            obj.metadata.deletion_timestamp = None
+            self._objects[key] = obj
+            events.append((ADDED, obj))
+        self._dispatch(events)
+        return serde.deep_copy(obj)
+
+    def get(self, kind: str, namespace: str, name: str) -> Any:
+        with self._lock:
+            obj = self._objects.get((kind, namespace, name))
+            if obj is None:
+                raise NotFoundError(f"{kind} {namespace}/{name}")
+            return serde.deep_copy(obj)
+
+    def try_get(self, kind: str, namespace: str, name: str) -> Optional[Any]:
+        try:
+            return self.get(kind, namespace, name)
+        except NotFoundError:
+            return None
+
+    def list(self, kind: str, namespace: Optional[str] = None,
+             label_selector: Optional[dict[str, str]] = None,
+             filter_fn: Optional[Callable[[Any], bool]] = None) -> list[Any]:
+        out = []
+        with self._lock:
+            for (k, ns, _), obj in self._objects.items():
+                if k != kind:
+                    continue
+                if namespace is not None and ns != namespace:
+                    continue
+                if label_selector is not None:
+                    labels = obj.metadata.labels or {}
+                    if any(labels.get(lk) != lv for lk, lv in label_selector.items()):
+                        continue
+                if filter_fn is not None and not filter_fn(obj):
+                    continue
+                out.append(serde.deep_copy(obj))
+        out.sort(key=lambda o: (o.metadata.namespace, o.metadata.name))
+        return out
+
+    def _admit_update(self, obj: Any, old: Any) -> None:
+        kind = obj_kind(obj)
+        for fn in self._mutators.get(kind, []):
+            fn(obj)
+        for fn in self._validators.get(kind, []):
+            fn(obj, old)
+
+    def update(self, obj: Any) -> Any:
+        """Full update (spec+metadata). Bumps generation iff spec changed."""
+        obj = serde.deep_copy(obj)
+        events: list[tuple[str, Any]] = []
+        with self._lock:
+            key = obj_key(obj)
+            old = self._objects.get(key)
+            if old is None:
+                raise NotFoundError(f"{key}")
+            if obj.metadata.resource_version and \
+                    obj.metadata.resource_version != old.metadata.resource_version:
+                raise ConflictError(f"{key}: stale resourceVersion")
+            self._admit_update(obj, serde.deep_copy(old))
+            self._rv += 1
+            obj.metadata.uid = old.metadata.uid
+            obj.metadata.creation_timestamp = old.metadata.creation_timestamp
+            obj.metadata.deletion_timestamp = old.metadata.deletion_timestamp
+            obj.metadata.resource_version = str(self._rv)
+            obj.metadata.generation = old.metadata.generation
+            if hasattr(obj, "spec") and \
+                    serde.to_dict(obj.spec) != serde.to_dict(old.spec):
+                obj.metadata.generation += 1
+            # status is updated through update_status only
+            if hasattr(obj, "status"):
+                obj.status = serde.deep_copy(old.status)
+            self._objects[key] = obj
+            events.append((MODIFIED, obj))
+        self._dispatch(events)
+        self._maybe_finish_foreground_owners()
+        return serde.deep_copy(obj)
+
+    def update_status(self, obj: Any) -> Any:
+        obj = serde.deep_copy(obj)
+        events: list[tuple[str, Any]] = []
+        with self._lock:
+            key = obj_key(obj)
+            old = self._objects.get(key)
+            if old is None:
+                raise NotFoundError(f"{key}")
+            if obj.metadata.resource_version and \
+                    obj.metadata.resource_version != old.metadata.resource_version:
+                raise ConflictError(f"{key}: stale resourceVersion")
+            self._rv += 1
+            stored = serde.deep_copy(old)
+            stored.status = serde.deep_copy(obj.status)
+            stored.metadata.resource_version = str(self._rv)
+            self._objects[key] = stored
+            events.append((MODIFIED, stored))
+        self._dispatch(events)
+        return serde.deep_copy(stored)
+
+    def apply(self, obj: Any, field_manager: str = "lws") -> Any:
+        """Server-side-apply equivalent: create-or-take-ownership update of
+        spec, labels, annotations and ownerReferences (the reference uses
+        client.Apply with force ownership, fieldManager "lws" —
+        leaderworkerset_controller.go:381-417).  Status is preserved."""
+        with self._lock:
+            existing = self._objects.get(obj_key(obj))
+        if existing is None:
+            return self.create(obj)
+        new = serde.deep_copy(existing)
+        new.metadata.labels = dict(obj.metadata.labels or {})
+        new.metadata.annotations = dict(obj.metadata.annotations or {})
+        if obj.metadata.owner_references:
+            new.metadata.owner_references = serde.deep_copy(obj.metadata.owner_references)
+        if hasattr(obj, "spec"):
+            new.spec = serde.deep_copy(obj.spec)
+        return self.update(new)
+
+    # -- deletion / GC ------------------------------------------------------
+    def delete(self, kind: str, namespace: str, name: str,
+               propagation: str = "Background") -> None:
+        events: list[tuple[str, Any]] = []
+        to_cascade: list[tuple[str, str, str]] = []
+        with self._lock:
+            key = (kind, namespace, name)
+            obj = self._objects.get(key)
+            if obj is None:
+                raise NotFoundError(f"{key}")
+            if propagation == "Foreground":
+                if obj.metadata.deletion_timestamp is None:
+                    obj.metadata.deletion_timestamp = time.time()
+                    if FOREGROUND_FINALIZER not in obj.metadata.finalizers:
+                        obj.metadata.finalizers.append(FOREGROUND_FINALIZER)
+                    self._rv += 1
+                    obj.metadata.resource_version = str(self._rv)
+                    events.append((MODIFIED, obj))
+                for dep_key in self._dependents_locked(obj.metadata.uid):
+                    to_cascade.append(dep_key)
+            else:
+                if obj.metadata.deletion_timestamp is None and obj.metadata.finalizers:
+                    obj.metadata.deletion_timestamp = time.time()
+                    self._rv += 1
+                    obj.metadata.resource_version = str(self._rv)
+                    events.append((MODIFIED, obj))
+                elif not obj.metadata.finalizers:
+                    del self._objects[key]
+                    events.append((DELETED, obj))
+                    if propagation != "Orphan":
+                        for dep_key in self._dependents_locked(obj.metadata.uid):
+                            to_cascade.append(dep_key)
+        self._dispatch(events)
+        for dk, dns_, dn in to_cascade:
+            try:
+                # children of a foreground-deleted owner are themselves
+                # deleted foreground so the cascade is depth-first
+                self.delete(dk, dns_, dn,
+                            propagation="Foreground" if propagation == "Foreground"
+                            else "Background")
+            except NotFoundError:
+                pass
+        if propagation == "Foreground":
+            self._maybe_finish_foreground_owners()
+
+    def _dependents_locked(self, owner_uid: str) -> list[tuple[str, str, str]]:
+        return [k for k, o in self._objects.items()
+                if any(ref.uid == owner_uid for ref in o.metadata.owner_references)]
+
+    def remove_finalizer(self, kind: str, namespace: str, name: str,
+                         finalizer: str) -> None:
+        events: list[tuple[str, Any]] = []
+        with self._lock:
+            obj = self._objects.get((kind, namespace, name))
+            if obj is None:
+                return
+            if finalizer in obj.metadata.finalizers:
+                obj.metadata.finalizers.remove(finalizer)
+                self._rv += 1
+                obj.metadata.resource_version = str(self._rv)
+            if obj.metadata.deletion_timestamp is not None and \
+                    not [f for f in obj.metadata.finalizers if f != FOREGROUND_FINALIZER]:
+                pass
+        self._finalize_pending()
+        self._maybe_finish_foreground_owners()
+        self._dispatch(events)
+
+    def add_finalizer(self, kind: str, namespace: str, name: str,
+                      finalizer: str) -> None:
+        with self._lock:
+            obj = self._objects.get((kind, namespace, name))
+            if obj is None:
+                raise NotFoundError(f"{kind} {namespace}/{name}")
+            if finalizer not in obj.metadata.finalizers:
+                obj.metadata.finalizers.append(finalizer)
+                self._rv += 1
+                obj.metadata.resource_version = str(self._rv)
+
+    def _finalize_pending(self) -> None:
+        """Remove objects whose deletionTimestamp is set and whose only
+        remaining finalizer is (at most) the foreground one with no
+        dependents left."""
+        events: list[tuple[str, Any]] = []
+        with self._lock:
+            for key in list(self._objects.keys()):
+                obj = self._objects[key]
+                if obj.metadata.deletion_timestamp is None:
+                    continue
+                others = [f for f in obj.metadata.finalizers
+                          if f != FOREGROUND_FINALIZER]
+                if others:
+                    continue
+                if FOREGROUND_FINALIZER in obj.metadata.finalizers and \
+                        self._dependents_locked(obj.metadata.uid):
+                    continue
+                del self._objects[key]
+                events.append((DELETED, obj))
+        if events:
+            self._dispatch(events)
+            # removal may unblock a foreground parent
+            self._maybe_finish_foreground_owners()
+            # cascade to background dependents of removed objects
+            for _, obj in events:
+                for dk, dns_, dn in self._dependents(obj.metadata.uid):
+                    try:
+                        self.delete(dk, dns_, dn, propagation="Background")
+                    except NotFoundError:
+                        pass
+
+    def _dependents(self, owner_uid: str) -> list[tuple[str, str, str]]:
+        with self._lock:
+            return self._dependents_locked(owner_uid)
+
+    def _maybe_finish_foreground_owners(self) -> None:
+        self._finalize_pending()
+
+    # -- introspection ------------------------------------------------------
+    def snapshot_keys(self) -> list[tuple[str, str, str]]:
+        with self._lock:
+            return sorted(self._objects.keys())

@@ -1,0 +1,55 @@
+import pytest
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: tests that require an MI355X GPU (run via gpurun)")
+
+
+@pytest.fixture
+def cluster():
+    """A started single-node fake cluster (8 GPUs, FakeRuntime)."""
+    from lws_amd.cluster.cluster import LwsCluster, make_nodes
+
+    c = LwsCluster(nodes=make_nodes(1, gpus_per_node=8)).start()
+    yield c
+    c.stop()
+
+
+def make_lws(name="my-lws", namespace="default", replicas=1, size=2,
+             image="engine:latest", **kwargs):
+    """Builder-pattern fixture (reference test/wrappers/wrappers.go)."""
+    from lws_amd.api.core import Container, PodSpec, PodTemplateSpec
+    from lws_amd.api.leaderworkerset import (LeaderWorkerSet,
+                                             LeaderWorkerSetSpec,
+                                             LeaderWorkerTemplate)
+    from lws_amd.api.meta import ObjectMeta
+
+    lws = LeaderWorkerSet()
+    lws.metadata = ObjectMeta(name=name, namespace=namespace)
+    lws.spec = LeaderWorkerSetSpec(
+        replicas=replicas,
+        leader_worker_template=LeaderWorkerTemplate(
+            size=size,
+            worker_template=PodTemplateSpec(
+                spec=PodSpec(containers=[Container(name="main", image=image)]))),
+        **kwargs)
+    return lws
+
+
+def wait_for(fn, timeout=20.0, interval=0.02, desc="condition"):
+    import time
+    deadline = time.monotonic() + timeout
+    while time.monotonic() < deadline:
+        result = fn()
+        if result:
+            return result
+        time.sleep(interval)
+    raise AssertionError(f"timed out waiting for {desc}")
+
+
+def lws_condition(lws, cond_type):
+    for c in lws.status.conditions:
+        if c.type == cond_type:
+            return c
+    return None
