@@ -65,6 +65,7 @@ class Scheduler:
             node = self._find_node(p, scheduled)
             if node is not None:
                 self._bind(p, node)
+                p.node_name = node.metadata.name  # visible to later fit checks
                 scheduled.append(p)
                 progress = True
 

@@ -228,7 +228,10 @@ class StatefulSetController:
             if p.metadata.labels.get(STS_REVISION_LABEL) == current_rev)
         st.current_revision = current_rev
         st.update_revision = update_rev
-        self.store.update_status(fresh)
+        try:
+            self.store.update_status(fresh)
+        except NotFoundError:
+            pass  # sts deleted concurrently
 
 
 def statefulset_ready(sts: StatefulSet) -> bool:

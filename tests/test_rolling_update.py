@@ -1,0 +1,286 @@
+"""Rolling update + restart policy behavior tests.
+
+Mirrors the reference integration scenarios (rolling update with
+maxUnavailable/maxSurge, partition monotonicity, all-or-nothing restart —
+test/integration/controllers/leaderworkerset_test.go).
+"""
+import pytest
+
+from lws_amd.api import leaderworkerset as lwsapi
+from lws_amd.controllers.leaderworkerset_controller import (
+    calculate_continuous_ready_replicas, calculate_lws_unready_replicas,
+    calculate_rolling_update_replicas, rolling_update_partition)
+from tests.conftest import lws_condition, make_lws, wait_for
+
+
+# ---------------------------------------------------------------------------
+# pure-math unit tests (reference leaderworkerset_controller_test.go style)
+
+def test_rolling_update_partition_math():
+    R, U = True, False  # ready flags
+    # states: (ready, updated)
+    # all ready+old, step 1: partition moves to replicas-1
+    states = [(True, False)] * 4
+    assert rolling_update_partition(states, 4, 1, 4) == 3
+    # tail updated+ready → partition keeps walking down
+    states = [(True, False), (True, False), (True, True), (True, True)]
+    assert rolling_update_partition(states, 4, 1, 2) == 1
+    # not-ready updated tail → partition stays (waits for readiness)
+    states = [(True, False), (True, False), (True, False), (False, True)]
+    assert rolling_update_partition(states, 4, 1, 3) == 3
+    # stuck unblocking: all replicas not ready → partition advances to the
+    # rollingStepPartition floor (never below it) so the rollout can proceed
+    states = [(False, False)] * 4
+    assert rolling_update_partition(states, 4, 1, 4) == 3
+    # maxUnavailable accounting: one old replica below rollingStepPartition
+    # not ready → partition raised by 1 vs naive step
+    states = [(False, False), (True, False), (True, False), (True, True)]
+    # continuous ready tail = 1, step = 1 → rollingStepPartition = 2,
+    # unavailable below = 1 → partition = 3, then walk-down: idx3 updated →3
+    # idx2: ready+old → stop. => 3, capped by current partition
+    assert rolling_update_partition(states, 4, 1, 3) == 3
+    # monotonicity: never exceeds current partition
+    states = [(True, True)] * 4
+    assert rolling_update_partition(states, 4, 1, 0) == 0
+
+
+def test_calculate_rolling_update_replicas():
+    # no surge: always lwsReplicas
+    assert calculate_rolling_update_replicas(4, 0, 1, 4) == 4
+    # surge active while many unready
+    assert calculate_rolling_update_replicas(4, 2, 1, 4) == 6
+    # reclaim: unready within surge → keep only required surplus
+    assert calculate_rolling_update_replicas(4, 2, 1, 2) == 5
+    assert calculate_rolling_update_replicas(4, 2, 1, 1) == 4
+    assert calculate_rolling_update_replicas(4, 2, 1, 0) == 4
+
+
+def test_continuous_ready_and_unready():
+    states = [(False, False), (True, True), (True, True)]
+    assert calculate_continuous_ready_replicas(states) == 2
+    assert calculate_lws_unready_replicas(states, 3) == 1
+    assert calculate_lws_unready_replicas([], 3) == 3
+
+
+# ---------------------------------------------------------------------------
+# cluster behavior
+
+def _wait_available(cluster, name, ns="default", timeout=30):
+    def available():
+        cur = cluster.get_lws(ns, name)
+        cond = lws_condition(cur, "Available")
+        return cur if cond is not None and cond.status == "True" else None
+    return wait_for(available, desc=f"{name} Available", timeout=timeout)
+
+
+def test_rolling_update_replaces_all_groups(cluster):
+    lws = make_lws(name="roll", replicas=3, size=2)
+    cluster.store.create(lws)
+    _wait_available(cluster, "roll")
+
+    pods_before = {p.metadata.name: p.metadata.uid
+                   for p in cluster.store.list("Pod", "default")}
+    assert len(pods_before) == 6
+
+    cur = cluster.get_lws("default", "roll")
+    cur.spec.leader_worker_template.worker_template.spec.containers[0].image = \
+        "engine:v2"
+    cluster.store.update(cur)
+
+    # UpdateInProgress should appear
+    def updating():
+        c = cluster.get_lws("default", "roll")
+        cond = lws_condition(c, "UpdateInProgress")
+        return c if cond is not None and cond.status == "True" else None
+    wait_for(updating, desc="UpdateInProgress", timeout=10)
+
+    def done():
+        c = cluster.get_lws("default", "roll")
+        cond_a = lws_condition(c, "Available")
+        cond_u = lws_condition(c, "UpdateInProgress")
+        return (c if cond_a is not None and cond_a.status == "True"
+                and c.status.updated_replicas == 3 and cond_u is None
+                else None)
+    wait_for(done, desc="rolling update complete", timeout=60)
+
+    # every group pod replaced, new image everywhere
+    pods_after = cluster.store.list("Pod", "default")
+    assert len(pods_after) == 6
+    for p in pods_after:
+        assert p.metadata.uid not in pods_before.values()
+        assert p.spec.containers[0].image == "engine:v2"
+
+    # revisions truncated to the current one (happens right after the final
+    # status write in the same reconcile — poll briefly)
+    wait_for(lambda: len(cluster.store.list(
+        "ControllerRevision", "default",
+        label_selector={lwsapi.SET_NAME_LABEL_KEY: "roll"})) == 1,
+        desc="revision truncation", timeout=10)
+
+
+def test_rolling_update_with_max_surge(cluster):
+    lws = make_lws(name="surge", replicas=2, size=2)
+    from lws_amd.api.leaderworkerset import (RollingUpdateConfiguration,
+                                             RolloutStrategy)
+    lws.spec.rollout_strategy = RolloutStrategy(
+        type="RollingUpdate",
+        rolling_update_configuration=RollingUpdateConfiguration(
+            partition=0, max_unavailable=1, max_surge=1))
+    cluster.store.create(lws)
+    _wait_available(cluster, "surge")
+
+    cur = cluster.get_lws("default", "surge")
+    cur.spec.leader_worker_template.worker_template.spec.containers[0].image = \
+        "engine:v2"
+    cluster.store.update(cur)
+
+    def done():
+        c = cluster.get_lws("default", "surge")
+        cond = lws_condition(c, "Available")
+        sts = cluster.store.try_get("StatefulSet", "default", "surge")
+        live = [p for p in cluster.store.list("Pod", "default")
+                if p.metadata.deletion_timestamp is None]
+        return (c if cond is not None and cond.status == "True"
+                and c.status.updated_replicas == 2
+                and sts.spec.replicas == 2 and len(live) == 4
+                and all(p.spec.containers[0].image == "engine:v2"
+                        for p in live) else None)
+    wait_for(done, desc="surge rollout complete + reclaimed", timeout=60)
+
+
+def test_partition_blocks_lower_ordinals(cluster):
+    from lws_amd.api.leaderworkerset import (RollingUpdateConfiguration,
+                                             RolloutStrategy)
+    lws = make_lws(name="part", replicas=3, size=2)
+    cluster.store.create(lws)
+    _wait_available(cluster, "part")
+
+    cur = cluster.get_lws("default", "part")
+    cur.spec.rollout_strategy.rolling_update_configuration.partition = 2
+    cur.spec.leader_worker_template.worker_template.spec.containers[0].image = \
+        "engine:v2"
+    cluster.store.update(cur)
+
+    # only group 2 updates; groups 0,1 stay on v1
+    def partial():
+        c = cluster.get_lws("default", "part")
+        return c if c.status.updated_replicas == 1 and \
+            c.status.ready_replicas == 3 else None
+    wait_for(partial, desc="partition-limited update", timeout=60)
+    import time
+    time.sleep(0.3)
+    by_name = {p.metadata.name: p for p in cluster.store.list("Pod", "default")}
+    assert by_name["part-0"].spec.containers[0].image == "engine:latest"
+    assert by_name["part-1"].spec.containers[0].image == "engine:latest"
+    assert by_name["part-2"].spec.containers[0].image == "engine:v2"
+
+    # release the partition → full rollout
+    cur = cluster.get_lws("default", "part")
+    cur.spec.rollout_strategy.rolling_update_configuration.partition = 0
+    cluster.store.update(cur)
+
+    def done():
+        c = cluster.get_lws("default", "part")
+        cond = lws_condition(c, "Available")
+        return c if cond is not None and cond.status == "True" and \
+            c.status.updated_replicas == 3 else None
+    wait_for(done, desc="full rollout after partition release", timeout=60)
+
+
+def test_restart_policy_recreates_group(cluster):
+    lws = make_lws(name="restart", replicas=1, size=3)
+    cluster.store.create(lws)
+    _wait_available(cluster, "restart")
+
+    pods = cluster.store.list("Pod", "default")
+    uids_before = {p.metadata.name: p.metadata.uid for p in pods}
+    assert len(pods) == 3
+
+    # simulate container restart on a worker pod
+    worker = next(p for p in pods if p.metadata.name == "restart-0-2")
+    agent = cluster.agents[0]
+    agent.mark_container_restarted(worker)
+
+    # whole group must be recreated (new UIDs for every pod)
+    def recreated():
+        cur_pods = cluster.store.list("Pod", "default")
+        if len(cur_pods) != 3:
+            return None
+        cur = {p.metadata.name: p.metadata.uid for p in cur_pods}
+        if set(cur) != set(uids_before):
+            return None
+        if any(cur[n] == uids_before[n] for n in cur):
+            return None
+        return cur_pods
+    wait_for(recreated, desc="group recreated with new pods", timeout=30)
+    _wait_available(cluster, "restart")
+
+
+def test_none_restart_policy_keeps_group(cluster):
+    lws = make_lws(name="keep", replicas=1, size=2)
+    lws.spec.leader_worker_template.restart_policy = "None"
+    cluster.store.create(lws)
+    _wait_available(cluster, "keep")
+    pods = cluster.store.list("Pod", "default")
+    uids_before = {p.metadata.name: p.metadata.uid for p in pods}
+
+    worker = next(p for p in pods if p.metadata.name == "keep-0-1")
+    cluster.agents[0].mark_container_restarted(worker)
+    import time
+    time.sleep(0.5)
+    cur = {p.metadata.name: p.metadata.uid
+           for p in cluster.store.list("Pod", "default")}
+    assert cur == uids_before  # nothing recreated
+
+
+def test_leader_ready_startup_policy():
+    from lws_amd.cluster.cluster import LwsCluster, make_nodes
+    from lws_amd.cluster.node import FakeRuntime
+
+    c = LwsCluster(nodes=make_nodes(1, gpus_per_node=8),
+                   runtime_factory=lambda n: FakeRuntime(ready_delay=0.4)).start()
+    try:
+        lws = make_lws(name="lr", replicas=1, size=2,
+                       startup_policy="LeaderReady")
+        c.store.create(lws)
+        # while the leader is not ready, no worker STS may exist
+        wait_for(lambda: c.store.try_get("Pod", "default", "lr-0"),
+                 desc="leader pod")
+        assert c.store.try_get("StatefulSet", "default", "lr-0") is None
+        wait_for(lambda: c.store.try_get("StatefulSet", "default", "lr-0"),
+                 desc="worker sts after leader ready", timeout=20)
+    finally:
+        c.stop()
+
+
+def test_exclusive_topology_placement():
+    from lws_amd.cluster.cluster import LwsCluster, make_nodes
+
+    topo = "topology.lws.amd.com/island"
+    # 4 nodes with 2 GPUs each; groups of size 2 with 1 GPU/pod must land
+    # 1:1 on islands
+    c = LwsCluster(nodes=make_nodes(4, gpus_per_node=2)).start()
+    try:
+        lws = make_lws(name="excl", replicas=4, size=2)
+        lws.metadata.annotations = {lwsapi.EXCLUSIVE_KEY_ANNOTATION_KEY: topo}
+        for tmpl in [lws.spec.leader_worker_template.worker_template]:
+            tmpl.spec.containers[0].resources.requests = {"amd.com/gpu": 1}
+        c.store.create(lws)
+
+        def all_scheduled():
+            pods = c.store.list("Pod", "default")
+            if len(pods) != 8 or any(not p.node_name for p in pods):
+                return None
+            return pods
+        pods = wait_for(all_scheduled, desc="8 pods scheduled", timeout=30)
+        # each group on exactly one island; no island shared between groups
+        groups = {}
+        for p in pods:
+            g = p.metadata.labels[lwsapi.GROUP_INDEX_LABEL_KEY]
+            node = c.node(p.node_name)
+            groups.setdefault(g, set()).add(node.metadata.labels[topo])
+        assert all(len(islands) == 1 for islands in groups.values())
+        all_islands = [next(iter(v)) for v in groups.values()]
+        assert len(set(all_islands)) == 4
+    finally:
+        c.stop()
