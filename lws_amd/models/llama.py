@@ -1,0 +1,336 @@
+"""Llama-family model, MI355X-native, TP-sharded.
+
+The serving data plane of lws_amd: on GPU the hot
+ops are the in-repo HIP/CDNA4 kernels (rmsnorm, fused residual add, rope,
+silu-mul, paged GQA decode attention); plain GEMMs go to hipBLASLt via
+torch.matmul; TP all-reduce rides RCCL/xGMI.  The CPU path (used only by
+the CPU test suite — never on a GPU host) runs the fp32 reference ops.
+
+Config presets follow the public Llama-3 architecture dimensions
+(BASELINE.json names Llama-3-8B and Llama-3-70B TP=8 as the bench models;
+weights are always random-init — there is no network for checkpoints).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+from ..parallel import tp as tpmod
+from ..parallel.tp import ShardedLinear, all_gather_cat, all_reduce
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama"
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_q_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    vocab_size: int = 128256
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+    max_position: int = 8192
+
+    @property
+    def q_size(self) -> int:
+        return self.num_q_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.num_kv_heads * self.head_dim
+
+    def num_params(self) -> int:
+        h, i, v = self.hidden_size, self.intermediate_size, self.vocab_size
+        per_layer = (h * self.q_size + 2 * h * self.kv_size + self.q_size * h
+                     + 3 * h * i + 2 * h)
+        return self.num_layers * per_layer + 2 * v * h + h
+
+
+def llama3_8b() -> LlamaConfig:
+    return LlamaConfig(name="llama-3-8b", hidden_size=4096,
+                       intermediate_size=14336, num_layers=32, num_q_heads=32,
+                       num_kv_heads=8, vocab_size=128256)
+
+
+def llama3_70b() -> LlamaConfig:
+    return LlamaConfig(name="llama-3-70b", hidden_size=8192,
+                       intermediate_size=28672, num_layers=80, num_q_heads=64,
+                       num_kv_heads=8, vocab_size=128256)
+
+
+def llama_tiny() -> LlamaConfig:
+    """Small config for tests/smoke (structure-identical to Llama-3)."""
+    return LlamaConfig(name="llama-tiny", hidden_size=256,
+                       intermediate_size=512, num_layers=2, num_q_heads=8,
+                       num_kv_heads=2, head_dim=128, vocab_size=1024,
+                       max_position=2048)
+
+
+MODEL_PRESETS = {
+    "llama-3-8b": llama3_8b,
+    "llama-3-70b": llama3_70b,
+    "llama-tiny": llama_tiny,
+}
+
+
+# ---------------------------------------------------------------------------
+# batch metadata
+
+@dataclass
+class PrefillBatch:
+    """Variable-length prompt batch, flattened to [T, ...]."""
+    input_ids: torch.Tensor          # [T] int64
+    positions: torch.Tensor          # [T] int32
+    seq_starts: list[int]            # len B+1 prefix offsets into T
+    slot_mapping: torch.Tensor       # [T] int64 cache slots
+
+
+@dataclass
+class DecodeBatch:
+    input_ids: torch.Tensor          # [B] int64
+    positions: torch.Tensor          # [B] int32
+    block_tables: torch.Tensor       # [B, max_pages] int32
+    seq_lens: torch.Tensor           # [B] int32 (incl. the new token)
+    slot_mapping: torch.Tensor       # [B] int64
+
+
+class _Ops:
+    """Dispatch shim: HIP kernels on GPU (mandatory), fp32 reference on CPU
+    (test-only).  On a CUDA device the native extension is REQUIRED — there
+    is no eager fallback."""
+
+    def __init__(self, device: torch.device):
+        self.is_gpu = device.type == "cuda"
+        if self.is_gpu:
+            import lws_amd.ops as ops
+            ops.require_native()
+            self.ops = ops
+        else:
+            from lws_amd.ops import build_rope_table  # table is host-side
+            import lws_amd.ops.reference as ref
+            self.ref = ref
+
+
+# ---------------------------------------------------------------------------
+
+class LlamaLayer:
+    def __init__(self, cfg: LlamaConfig, tp_rank: int, tp_world: int, device,
+                 dtype=torch.bfloat16):
+        h, d = cfg.hidden_size, cfg.head_dim
+        self.cfg = cfg
+        self.tp_world = tp_world
+        self.hq = cfg.num_q_heads // tp_world
+        self.hkv = max(1, cfg.num_kv_heads // tp_world)
+        assert cfg.num_q_heads % tp_world == 0, "q heads must divide TP"
+        assert cfg.num_kv_heads % tp_world == 0 or tp_world <= cfg.num_kv_heads
+        self.qkv = ShardedLinear(cfg.q_size + 2 * cfg.kv_size, h, 0,
+                                 tp_rank, tp_world, device, dtype)
+        self.o = ShardedLinear(h, cfg.q_size, 1, tp_rank, tp_world, device, dtype)
+        self.gate_up = ShardedLinear(2 * cfg.intermediate_size, h, 0,
+                                     tp_rank, tp_world, device, dtype)
+        self.down = ShardedLinear(h, cfg.intermediate_size, 1, tp_rank,
+                                  tp_world, device, dtype)
+        self.input_norm = torch.empty(h, device=device, dtype=dtype)
+        self.post_norm = torch.empty(h, device=device, dtype=dtype)
+        self.q_slice = self.hq * d
+        self.kv_slice = self.hkv * d
+
+    def materialize(self, gen=None):
+        for lin in (self.qkv, self.o, self.gate_up, self.down):
+            lin.materialize(gen)
+        self.input_norm.fill_(1.0)
+        self.post_norm.fill_(1.0)
+
+
+class LlamaForCausalLM:
+    def __init__(self, cfg: LlamaConfig, tp_rank: int = 0, tp_world: int = 1,
+                 device: str = "cpu", dtype=torch.bfloat16):
+        self.cfg = cfg
+        self.tp_rank = tp_rank
+        self.tp_world = tp_world
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self._ops = _Ops(self.device)
+        self.layers = [LlamaLayer(cfg, tp_rank, tp_world, self.device, dtype)
+                       for _ in range(cfg.num_layers)]
+        self.embed = torch.empty(cfg.vocab_size, cfg.hidden_size,
+                                 device=self.device, dtype=dtype)
+        self.final_norm = torch.empty(cfg.hidden_size, device=self.device,
+                                      dtype=dtype)
+        self.lm_head = ShardedLinear(cfg.vocab_size, cfg.hidden_size, 0,
+                                     tp_rank, tp_world, self.device, dtype)
+        from lws_amd.ops import build_rope_table
+        self.rope_table = build_rope_table(cfg.max_position, cfg.head_dim,
+                                           cfg.rope_theta, device=self.device)
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+
+    # -- weights --------------------------------------------------------
+    def materialize(self, seed: int = 0) -> int:
+        """Random-init all shards on-device; returns local param count."""
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed + self.tp_rank)
+        n = 0
+        for layer in self.layers:
+            layer.materialize(gen)
+            n += sum(l.numel for l in (layer.qkv, layer.o, layer.gate_up,
+                                       layer.down))
+            n += layer.input_norm.numel() + layer.post_norm.numel()
+        self.embed.normal_(0.0, 0.02, generator=gen)
+        self.final_norm.fill_(1.0)
+        self.lm_head.materialize(gen)
+        n += self.embed.numel() + self.final_norm.numel() + self.lm_head.numel
+        return n
+
+    # -- kv cache shape -------------------------------------------------
+    def kv_cache_spec(self) -> tuple[int, int]:
+        """(num_layers, kv_heads_per_rank)"""
+        return self.cfg.num_layers, self.layers[0].hkv
+
+    # -- core ops (GPU: HIP kernels; CPU: fp32 reference) ---------------
+    def _rmsnorm(self, x, w):
+        if self._ops.is_gpu:
+            return self._ops.ops.rmsnorm(x, w, self.cfg.rms_eps)
+        return self._ops.ref.rmsnorm_ref(x, w, self.cfg.rms_eps)
+
+    def _fused_add_rmsnorm(self, x, residual, w):
+        if self._ops.is_gpu:
+            self._ops.ops.fused_add_rmsnorm(x, residual, w, self.cfg.rms_eps)
+            return x, residual
+        out, new_res = self._ops.ref.fused_add_rmsnorm_ref(
+            x, residual, w, self.cfg.rms_eps)
+        return out, new_res
+
+    def _rope(self, q, k, positions, hq, hkv):
+        if self._ops.is_gpu:
+            self._ops.ops.rope(q, k, self.rope_table, positions, hq, hkv)
+            return q, k
+        return self._ops.ref.rope_ref(q, k, self.rope_table, positions, hq, hkv)
+
+    def _silu_mul(self, gu):
+        if self._ops.is_gpu:
+            return self._ops.ops.silu_mul(gu)
+        return self._ops.ref.silu_mul_ref(gu)
+
+    def _write_cache(self, k, v, k_cache, v_cache, slot_mapping):
+        if self._ops.is_gpu:
+            self._ops.ops.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+            return
+        page = k_cache.size(2)
+        for t in range(k.size(0)):
+            s = int(slot_mapping[t])
+            if s < 0:
+                continue
+            k_cache[s // page, :, s % page] = k[t]
+            v_cache[s // page, :, s % page] = v[t]
+
+    # -- forward --------------------------------------------------------
+    def forward_prefill(self, batch: PrefillBatch, kv_caches) -> torch.Tensor:
+        """Returns hidden states of the LAST token of each sequence [B, H]."""
+        cfg = self.cfg
+        x = self.embed[batch.input_ids]           # [T, H] (replicated embed)
+        residual = None
+        for li, layer in enumerate(self.layers):
+            if residual is None:
+                residual = x
+                h = self._rmsnorm(x, layer.input_norm)
+            else:
+                h, residual = self._fused_add_rmsnorm(x, residual,
+                                                      layer.input_norm)
+            qkv = layer.qkv(h)
+            q, k, v = qkv.split([layer.q_slice, layer.kv_slice,
+                                 layer.kv_slice], dim=-1)
+            q = q.contiguous()
+            k = k.contiguous()
+            q, k = self._rope(q, k, batch.positions, layer.hq, layer.hkv)
+            k_cache, v_cache = kv_caches[li]
+            self._write_cache(k.view(-1, layer.hkv, cfg.head_dim),
+                              v.contiguous().view(-1, layer.hkv, cfg.head_dim),
+                              k_cache, v_cache, batch.slot_mapping)
+            attn = self._prefill_attention(q, k, v, batch, layer)
+            o = layer.o(attn)
+            x = all_reduce(o)
+            # MLP
+            h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)
+            gu = layer.gate_up(h)
+            act = self._silu_mul(gu)
+            x = all_reduce(layer.down(act))
+        h, _ = self._fused_add_rmsnorm(x, residual, self.final_norm)
+        last = torch.tensor([s - 1 for s in batch.seq_starts[1:]],
+                            device=h.device, dtype=torch.long)
+        return h[last]
+
+    def _prefill_attention(self, q, k, v, batch: PrefillBatch, layer):
+        """Per-sequence causal attention (library-GEMM composition in bf16
+        with fp32 softmax; the decode path is the hot loop and uses the HIP
+        paged-attention kernel)."""
+        cfg = self.cfg
+        d = cfg.head_dim
+        T = q.size(0)
+        out = torch.empty(T, layer.hq * d, device=q.device, dtype=q.dtype)
+        G = layer.hq // layer.hkv
+        for i in range(len(batch.seq_starts) - 1):
+            s0, s1 = batch.seq_starts[i], batch.seq_starts[i + 1]
+            S = s1 - s0
+            qs = q[s0:s1].view(S, layer.hq, d).transpose(0, 1)      # [Hq,S,D]
+            ks = k[s0:s1].view(S, layer.hkv, d).transpose(0, 1)
+            vs = v[s0:s1].view(S, layer.hkv, d).transpose(0, 1)
+            ks = ks.repeat_interleave(G, dim=0)
+            vs = vs.repeat_interleave(G, dim=0)
+            scores = (qs @ ks.transpose(-1, -2)).float() * self.scale
+            mask = torch.full((S, S), float("-inf"),
+                              device=q.device).triu_(1)
+            p = torch.softmax(scores + mask, dim=-1).to(q.dtype)
+            o = (p @ vs).transpose(0, 1).reshape(S, layer.hq * d)
+            out[s0:s1] = o
+        return out
+
+    def forward_decode(self, batch: DecodeBatch, kv_caches) -> torch.Tensor:
+        """One token per sequence: returns hidden states [B, H]."""
+        cfg = self.cfg
+        x = self.embed[batch.input_ids]           # [B, H]
+        residual = None
+        for li, layer in enumerate(self.layers):
+            if residual is None:
+                residual = x
+                h = self._rmsnorm(x, layer.input_norm)
+            else:
+                h, residual = self._fused_add_rmsnorm(x, residual,
+                                                      layer.input_norm)
+            qkv = layer.qkv(h)
+            q, k, v = qkv.split([layer.q_slice, layer.kv_slice,
+                                 layer.kv_slice], dim=-1)
+            q = q.contiguous()
+            k = k.contiguous()
+            q, k = self._rope(q, k, batch.positions, layer.hq, layer.hkv)
+            k_cache, v_cache = kv_caches[li]
+            self._write_cache(k.view(-1, layer.hkv, cfg.head_dim),
+                              v.contiguous().view(-1, layer.hkv, cfg.head_dim),
+                              k_cache, v_cache, batch.slot_mapping)
+            B = q.size(0)
+            qh = q.view(B, layer.hq, cfg.head_dim)
+            if self._ops.is_gpu:
+                attn = self._ops.ops.paged_attention_decode(
+                    qh, k_cache, v_cache, batch.block_tables, batch.seq_lens,
+                    self.scale)
+            else:
+                attn = self._ops.ref.paged_attention_decode_ref(
+                    qh, k_cache, v_cache, batch.block_tables, batch.seq_lens,
+                    self.scale)
+            o = layer.o(attn.view(B, layer.hq * cfg.head_dim))
+            x = all_reduce(o)
+            h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)
+            gu = layer.gate_up(h)
+            act = self._silu_mul(gu)
+            x = all_reduce(layer.down(act))
+        h, _ = self._fused_add_rmsnorm(x, residual, self.final_norm)
+        return h
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        """[N, H] -> [N, vocab] (column-sharded lm_head + all-gather)."""
+        local = self.lm_head(hidden)
+        return all_gather_cat(local, dim=-1)

@@ -1,0 +1,140 @@
+"""lws_amd.ops — hand-written CDNA4 (gfx950) kernels for the served engine.
+
+The GPU compute path is the HIP extension built in-tree by
+``lws_amd.ops.build``; there is NO eager/PyTorch fallback on GPU — if the
+extension is missing on a GPU host the ops raise, loudly, so a silently
+slow path can never masquerade as the native one.  The pure-PyTorch fp32
+implementations in ``lws_amd.ops.reference`` exist solely as numerics
+references for tests and for CPU-only development.
+"""
+from __future__ import annotations
+
+import ctypes
+import math
+from pathlib import Path
+from typing import Optional
+
+import torch
+
+_LIB = None
+_LOAD_ERROR: Optional[str] = None
+
+
+def _try_load() -> None:
+    global _LIB, _LOAD_ERROR
+    if _LIB is not None or _LOAD_ERROR is not None:
+        return
+    so = Path(__file__).resolve().parent / "_C.so"
+    if not so.exists():
+        _LOAD_ERROR = f"{so} not built (run python -m lws_amd.ops.build)"
+        return
+    try:
+        torch.ops.load_library(str(so))  # registers nothing; pybind module
+    except Exception:
+        pass
+    try:
+        # the extension is a plain pybind11 module compiled as _C.so;
+        # load it via importlib machinery
+        import importlib.util
+
+        spec = importlib.util.spec_from_file_location("lws_amd_C", str(so))
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)
+        _LIB = mod
+    except Exception as e:  # noqa: BLE001
+        _LOAD_ERROR = f"failed to load {so}: {e}"
+
+
+def native_available() -> bool:
+    _try_load()
+    return _LIB is not None
+
+
+def require_native():
+    _try_load()
+    if _LIB is None:
+        raise RuntimeError(
+            "lws_amd native kernels unavailable on a GPU host: "
+            f"{_LOAD_ERROR}. Build with `python -m lws_amd.ops.build`.")
+    return _LIB
+
+
+# ---------------------------------------------------------------------------
+# public op surface (GPU -> HIP kernels, loud failure if missing)
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5,
+            out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    lib = require_native()
+    if out is None:
+        out = torch.empty_like(x)
+    lib.rmsnorm(out, x, weight, eps)
+    return out
+
+
+def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float = 1e-5) -> None:
+    """In place: residual += x; x = rmsnorm(residual) * weight."""
+    lib = require_native()
+    lib.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+def silu_mul(gateup: torch.Tensor,
+             out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    lib = require_native()
+    shape = list(gateup.shape)
+    shape[-1] //= 2
+    if out is None:
+        out = torch.empty(shape, dtype=gateup.dtype, device=gateup.device)
+    lib.silu_mul(out, gateup)
+    return out
+
+
+def rope(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
+         positions: torch.Tensor, num_q_heads: int, num_kv_heads: int) -> None:
+    """In-place neox-style rotary on q [T, Hq*D] and k [T, Hkv*D]."""
+    lib = require_native()
+    lib.rope(q, k, cos_sin, positions, num_q_heads, num_kv_heads)
+
+
+def build_rope_table(max_pos: int, head_dim: int, theta: float = 10000.0,
+                     device="cpu") -> torch.Tensor:
+    """Host-precomputed fp32 [max_pos, head_dim] table: cos | sin halves."""
+    half = head_dim // 2
+    inv_freq = 1.0 / (theta ** (torch.arange(0, half, dtype=torch.float64) / half))
+    pos = torch.arange(max_pos, dtype=torch.float64)
+    ang = torch.outer(pos, inv_freq)
+    table = torch.cat([torch.cos(ang), torch.sin(ang)], dim=-1).float()
+    return table.to(device)
+
+
+def paged_attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                           v_cache: torch.Tensor, block_tables: torch.Tensor,
+                           seq_lens: torch.Tensor, scale: float,
+                           chunk_keys: int = 512,
+                           workspace: Optional[tuple] = None,
+                           out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    lib = require_native()
+    B, Hq, D = q.shape
+    Hkv = k_cache.size(1)
+    G = Hq // Hkv
+    max_len = int(block_tables.size(1)) * int(k_cache.size(2))
+    num_chunks = max(1, (max_len + chunk_keys - 1) // chunk_keys)
+    if workspace is None:
+        ws_acc = torch.empty((B, Hkv, num_chunks, G, D), dtype=torch.float32,
+                             device=q.device)
+        ws_ml = torch.empty((B, Hkv, num_chunks, G, 2), dtype=torch.float32,
+                            device=q.device)
+    else:
+        ws_acc, ws_ml = workspace
+        num_chunks = ws_ml.size(2)
+    if out is None:
+        out = torch.empty_like(q)
+    lib.paged_attention_decode(out, q, k_cache, v_cache, block_tables,
+                               seq_lens, ws_acc, ws_ml, scale, chunk_keys)
+    return out
+
+
+def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, slot_mapping: torch.Tensor) -> None:
+    lib = require_native()
+    lib.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)

@@ -1,0 +1,64 @@
+"""Build the lws_amd HIP kernel extension IN-TREE for gfx950.
+
+The built .so lands next to this file (lws_amd/ops/_C.so) so it travels to
+the GPU box with the repo snapshot (JIT caches under ~/.cache do not).
+hipcc cross-compiles gfx950 on CPU-only hosts, so this runs anywhere.
+"""
+from __future__ import annotations
+
+import os
+import shutil
+import subprocess
+import sys
+from pathlib import Path
+
+PKG_DIR = Path(__file__).resolve().parent
+CSRC = PKG_DIR / "csrc"
+SO_NAME = "_C.so"
+SOURCES = ["norm_act.hip", "rope.hip", "paged_attention.hip", "bindings.cpp"]
+
+
+def _newest_source_mtime() -> float:
+    return max((CSRC / s).stat().st_mtime for s in SOURCES) if CSRC.exists() else 0
+
+
+def needs_build() -> bool:
+    so = PKG_DIR / SO_NAME
+    if not so.exists():
+        return True
+    return so.stat().st_mtime < _newest_source_mtime()
+
+
+def build(verbose: bool = False, force: bool = False) -> Path:
+    so_path = PKG_DIR / SO_NAME
+    if not force and not needs_build():
+        return so_path
+
+    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    from torch.utils.cpp_extension import load
+
+    build_dir = PKG_DIR / "build"
+    build_dir.mkdir(exist_ok=True)
+    # torch's extension builder drives hipcc for .hip sources under ROCm and
+    # handles all torch/pybind include+link flags; the result is copied
+    # in-tree so gpurun ships it.
+    load(
+        name="lws_amd_C",
+        sources=[str(CSRC / s) for s in SOURCES],
+        extra_cflags=["-O3", "-std=c++17"],
+        extra_cuda_cflags=["-O3", "-std=c++17"],
+        build_directory=str(build_dir),
+        verbose=verbose,
+        is_python_module=True,
+        keep_intermediates=True,
+    )
+    built = build_dir / "lws_amd_C.so"
+    if not built.exists():
+        raise RuntimeError(f"extension build produced no {built}")
+    shutil.copy2(built, so_path)
+    return so_path
+
+
+if __name__ == "__main__":
+    p = build(verbose="-v" in sys.argv, force="-f" in sys.argv)
+    print(f"built {p}")

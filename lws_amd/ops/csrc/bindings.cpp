@@ -1,0 +1,31 @@
+// Python bindings for the lws_amd gfx950 kernel library.
+#include <torch/extension.h>
+
+void rmsnorm(torch::Tensor out, torch::Tensor input, torch::Tensor weight,
+             double eps);
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor weight, double eps);
+void silu_mul(torch::Tensor out, torch::Tensor gateup);
+void rope(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
+          torch::Tensor positions, long long num_q_heads,
+          long long num_kv_heads);
+void paged_attention_decode(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_tables, torch::Tensor seq_lens,
+                            torch::Tensor ws_acc, torch::Tensor ws_ml,
+                            double scale, long long chunk_keys);
+void reshape_and_cache(torch::Tensor k, torch::Tensor v,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor slot_mapping);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, gfx950)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+        "in-place residual add + RMSNorm (bf16, gfx950)");
+  m.def("silu_mul", &silu_mul, "fused SiLU-gate multiply (bf16, gfx950)");
+  m.def("rope", &rope, "fused rotary embedding for q,k (bf16, gfx950)");
+  m.def("paged_attention_decode", &paged_attention_decode,
+        "GQA paged attention decode with flash-decoding chunk split");
+  m.def("reshape_and_cache", &reshape_and_cache,
+        "scatter new k/v into the paged KV cache");
+}

@@ -1,0 +1,112 @@
+"""Tensor parallelism over RCCL/xGMI.
+
+One process per GPU; torch.distributed backend "nccl" IS RCCL on ROCm.
+The TP all-reduce after o_proj and down_proj (2/layer) rides the 7-link
+xGMI mesh.  Rendezvous comes from the env the lws_amd pod
+webhook injects (MASTER_ADDR/MASTER_PORT/WORLD_SIZE/NODE_RANK/
+LOCAL_WORLD_SIZE + LWS_* — lws_amd.accelerators.rccl).
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+
+@dataclass
+class ParallelState:
+    rank: int = 0
+    world_size: int = 1
+    local_rank: int = 0
+    device: torch.device = torch.device("cpu")
+    group: Optional[object] = None
+
+    @property
+    def is_distributed(self) -> bool:
+        return self.world_size > 1
+
+
+_STATE = ParallelState()
+
+
+def parallel_state() -> ParallelState:
+    return _STATE
+
+
+def init_distributed(backend: Optional[str] = None,
+                     device: Optional[str] = None) -> ParallelState:
+    """Initialize torch.distributed from the injected env (idempotent)."""
+    global _STATE
+    rank = int(os.environ.get("RANK", os.environ.get("NODE_RANK", "0")))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if world > 1 and not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+    if device is None:
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
+            device = f"cuda:{local_rank % torch.cuda.device_count()}"
+        else:
+            device = "cpu"
+    _STATE = ParallelState(rank=rank, world_size=world, local_rank=local_rank,
+                           device=torch.device(device))
+    return _STATE
+
+
+def all_reduce(t: torch.Tensor) -> torch.Tensor:
+    if _STATE.is_distributed and dist.is_initialized():
+        dist.all_reduce(t)
+    return t
+
+
+def all_gather_cat(t: torch.Tensor, dim: int = -1) -> torch.Tensor:
+    if not (_STATE.is_distributed and dist.is_initialized()):
+        return t
+    parts = [torch.empty_like(t) for _ in range(_STATE.world_size)]
+    dist.all_gather(parts, t.contiguous())
+    return torch.cat(parts, dim=dim)
+
+
+def barrier() -> None:
+    if _STATE.is_distributed and dist.is_initialized():
+        dist.barrier()
+
+
+# ---------------------------------------------------------------------------
+# sharded linear layers (bf16 weights, hipBLASLt GEMM via torch.matmul)
+
+class ShardedLinear:
+    """y = x @ W^T with W sharded along ``shard_dim`` across TP ranks.
+
+    shard_dim=0 (column-parallel): output features split; no comm.
+    shard_dim=1 (row-parallel): input features split; caller all-reduces.
+    """
+
+    def __init__(self, out_features: int, in_features: int, shard_dim: int,
+                 tp_rank: int, tp_world: int, device, dtype=torch.bfloat16):
+        self.shard_dim = shard_dim
+        if shard_dim == 0:
+            assert out_features % tp_world == 0
+            shape = (out_features // tp_world, in_features)
+        else:
+            assert in_features % tp_world == 0
+            shape = (out_features, in_features // tp_world)
+        self.weight = torch.empty(shape, device=device, dtype=dtype)
+
+    def materialize(self, generator: Optional[torch.Generator] = None,
+                    std: float = 0.02) -> None:
+        self.weight.normal_(0.0, std, generator=generator)
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        return x @ self.weight.t()
+
+    @property
+    def numel(self) -> int:
+        return self.weight.numel()

@@ -1,0 +1,178 @@
+"""GPU numerics tests: every HIP kernel vs a plain-PyTorch fp32 reference.
+
+All tests here are @pytest.mark.gpu and run on a real MI355X via gpurun.
+Tolerances account for bf16 I/O rounding (the kernels accumulate in fp32).
+"""
+import math
+
+import pytest
+import torch
+
+gpu = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs MI355X")
+
+
+def setup_module(module):
+    if torch.cuda.is_available():
+        import lws_amd.ops as ops
+        ops.require_native()  # fail loudly if the HIP extension is missing
+        torch.cuda.manual_seed(0)
+    torch.manual_seed(0)
+
+
+def assert_close_bf16(actual, expected, atol=2e-2, rtol=2e-2, msg=""):
+    actual = actual.float()
+    expected = expected.float()
+    torch.testing.assert_close(actual, expected, atol=atol, rtol=rtol, msg=msg)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("rows,D", [(1, 1024), (17, 4096), (4096, 8192),
+                                    (3, 128)])
+def test_rmsnorm(rows, D):
+    import lws_amd.ops as ops
+    from lws_amd.ops.reference import rmsnorm_ref
+
+    x = torch.randn(rows, D, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(D, dtype=torch.bfloat16, device="cuda")
+    out = ops.rmsnorm(x, w, eps=1e-5)
+    ref = rmsnorm_ref(x, w, eps=1e-5)
+    assert_close_bf16(out, ref)
+
+
+@gpu
+@requires_gpu
+def test_fused_add_rmsnorm():
+    import lws_amd.ops as ops
+    from lws_amd.ops.reference import fused_add_rmsnorm_ref
+
+    rows, D = 33, 8192
+    x = torch.randn(rows, D, dtype=torch.bfloat16, device="cuda")
+    res = torch.randn(rows, D, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(D, dtype=torch.bfloat16, device="cuda")
+    ref_norm, ref_res = fused_add_rmsnorm_ref(x, res, w, eps=1e-5)
+    ops.fused_add_rmsnorm(x, res, w, eps=1e-5)
+    assert_close_bf16(res, ref_res)
+    assert_close_bf16(x, ref_norm)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("rows,I", [(1, 1024), (65, 14336), (512, 3584)])
+def test_silu_mul(rows, I):
+    import lws_amd.ops as ops
+    from lws_amd.ops.reference import silu_mul_ref
+
+    gu = torch.randn(rows, 2 * I, dtype=torch.bfloat16, device="cuda")
+    out = ops.silu_mul(gu)
+    ref = silu_mul_ref(gu)
+    assert_close_bf16(out, ref)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("T,Hq,Hkv", [(7, 8, 1), (128, 8, 2), (1, 64, 8)])
+def test_rope(T, Hq, Hkv):
+    import lws_amd.ops as ops
+    from lws_amd.ops.reference import rope_ref
+
+    D = 128
+    table = ops.build_rope_table(4096, D, device="cuda")
+    q = torch.randn(T, Hq * D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, Hkv * D, dtype=torch.bfloat16, device="cuda")
+    pos = torch.randint(0, 4096, (T,), dtype=torch.int32, device="cuda")
+    ref_q, ref_k = rope_ref(q, k, table, pos, Hq, Hkv)
+    ops.rope(q, k, table, pos, Hq, Hkv)
+    assert_close_bf16(q, ref_q)
+    assert_close_bf16(k, ref_k)
+
+
+def _build_paged_cache(B, Hkv, max_len, page=16, device="cuda"):
+    max_pages_per_seq = (max_len + page - 1) // page
+    num_pages = B * max_pages_per_seq + 1
+    k_cache = torch.randn(num_pages, Hkv, page, 128, dtype=torch.bfloat16,
+                          device=device)
+    v_cache = torch.randn(num_pages, Hkv, page, 128, dtype=torch.bfloat16,
+                          device=device)
+    # shuffled page assignment to exercise the indirection
+    perm = torch.randperm(B * max_pages_per_seq, device=device) + 1
+    block_tables = perm.view(B, max_pages_per_seq).to(torch.int32)
+    return k_cache, v_cache, block_tables
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("B,Hq,Hkv,lens", [
+    (1, 8, 1, [5]),                # tiny, sub-page
+    (4, 8, 1, [16, 700, 33, 256]), # GQA 8 (Llama-70B TP8 shard shape)
+    (2, 16, 2, [1024, 2048]),      # multi-kv-head, long
+    (3, 4, 4, [8, 64, 100]),       # MHA (G=1)
+])
+def test_paged_attention_decode(B, Hq, Hkv, lens):
+    import lws_amd.ops as ops
+    from lws_amd.ops.reference import paged_attention_decode_ref
+
+    max_len = max(lens)
+    k_cache, v_cache, bt = _build_paged_cache(B, Hkv, max_len)
+    q = torch.randn(B, Hq, 128, dtype=torch.bfloat16, device="cuda")
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device="cuda")
+    scale = 1.0 / math.sqrt(128.0)
+    out = ops.paged_attention_decode(q, k_cache, v_cache, bt, seq_lens, scale)
+    ref = paged_attention_decode_ref(q, k_cache, v_cache, bt, seq_lens, scale)
+    assert_close_bf16(out, ref, atol=3e-2, rtol=3e-2)
+
+
+@gpu
+@requires_gpu
+def test_reshape_and_cache():
+    import lws_amd.ops as ops
+
+    T, Hkv, page = 37, 2, 16
+    num_pages = 16
+    k = torch.randn(T, Hkv, 128, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, Hkv, 128, dtype=torch.bfloat16, device="cuda")
+    k_cache = torch.zeros(num_pages, Hkv, page, 128, dtype=torch.bfloat16,
+                          device="cuda")
+    v_cache = torch.zeros_like(k_cache)
+    slots = torch.randperm(num_pages * page, device="cuda")[:T].to(torch.int64)
+    ops.reshape_and_cache(k, v, k_cache, v_cache, slots)
+    torch.cuda.synchronize()
+    for t in range(T):
+        s = int(slots[t])
+        pg, off = s // page, s % page
+        assert torch.equal(k_cache[pg, :, off], k[t])
+        assert torch.equal(v_cache[pg, :, off], v[t])
+
+
+@gpu
+@requires_gpu
+def test_decode_matches_prefill_math():
+    """Cross-check: paged decode on a 1-token query == full attention row."""
+    import lws_amd.ops as ops
+    from lws_amd.ops.reference import sdpa_prefill_ref
+
+    B, Hq, Hkv, S, D = 1, 8, 1, 129, 128
+    page = 16
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device="cuda")
+    q_full = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device="cuda")
+    scale = 1.0 / math.sqrt(D)
+    ref = sdpa_prefill_ref(q_full, k, v, scale, causal=True)[:, :, -1]  # [B,Hq,D]
+
+    # pack k/v into pages
+    npages = (S + page - 1) // page
+    k_cache = torch.zeros(npages + 1, Hkv, page, D, dtype=torch.bfloat16,
+                          device="cuda")
+    v_cache = torch.zeros_like(k_cache)
+    for p in range(npages):
+        n = min(page, S - p * page)
+        k_cache[p + 1, :, :n] = k[0, :, p * page:p * page + n]
+        v_cache[p + 1, :, :n] = v[0, :, p * page:p * page + n]
+    bt = (torch.arange(npages, device="cuda", dtype=torch.int32) + 1).view(1, -1)
+    seq_lens = torch.tensor([S], dtype=torch.int32, device="cuda")
+    out = ops.paged_attention_decode(q_full[:, :, -1].contiguous(), k_cache,
+                                     v_cache, bt, seq_lens, scale)
+    assert_close_bf16(out, ref, atol=3e-2, rtol=3e-2)
