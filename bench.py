@@ -1,0 +1,318 @@
+#!/usr/bin/env python3
+"""lws_amd flagship benchmark — BASELINE.json metric on MI355X.
+
+Measures **replica-group time-to-ready** and **rolling-update duration**
+for a 1-leader/(N-1)-worker LeaderWorkerSet serving Llama-3-70B TP=N
+(bf16, random-init weights, synthetic tokens), end to end through the real
+control plane: apply LWS CR -> webhooks -> leader STS -> scheduler -> node
+agents -> collective engine shard bring-up over RCCL/xGMI -> Available
+condition.  One bench "step" is one full lifecycle:
+
+    create LWS -> group Ready (t_ready)
+    template update -> rolling update -> group Ready on new revision
+        (t_rollout)
+    delete LWS -> pods drained, engine torn down
+
+Launch (the driver does this):
+    python bench.py --gpus N --steps K --warmup W
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 hosts the control plane + shard 0; other ranks host shards and
+follow broadcast commands (lws_amd/serving/runtime.py).  Rank 0 prints one
+JSON line.  scaling="strong": the model is fixed; more GPUs mean smaller
+shards per GPU.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+from lws_amd.api import leaderworkerset as lwsapi  # noqa: E402
+from lws_amd.serving.runtime import (BENCH_KV_PAGES_ANNOTATION,  # noqa: E402
+                                     BENCH_MODEL_ANNOTATION,
+                                     BENCH_SEED_ANNOTATION,
+                                     CollectiveGroupRuntime, Conductor,
+                                     WorkerLoop)
+
+METRIC = ("replica-group time-to-ready + rollout p50; "
+          "Llama-3-70B 1-leader/7-worker TP=8")
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--model", default="llama-3-70b")
+    p.add_argument("--device", default=None, help="cuda|cpu (auto)")
+    p.add_argument("--kv-pages", type=int, default=512)
+    p.add_argument("--decode-batch", type=int, default=32)
+    p.add_argument("--prompt-len", type=int, default=128)
+    p.add_argument("--decode-steps", type=int, default=32)
+    p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--skip-decode-bench", action="store_true")
+    return p.parse_args()
+
+
+def build_lws(args, size: int):
+    from lws_amd.api.core import Container, PodSpec, PodTemplateSpec, \
+        ResourceRequirements
+    from lws_amd.api.leaderworkerset import (LeaderWorkerSet,
+                                             LeaderWorkerSetSpec,
+                                             LeaderWorkerTemplate)
+    from lws_amd.api.meta import ObjectMeta
+
+    tmpl = PodTemplateSpec(
+        metadata=ObjectMeta(annotations={
+            BENCH_MODEL_ANNOTATION: args.model,
+            BENCH_KV_PAGES_ANNOTATION: str(args.kv_pages),
+            BENCH_SEED_ANNOTATION: str(args.seed),
+        }),
+        spec=PodSpec(containers=[Container(
+            name="engine", image="lws-amd-engine:bench",
+            resources=ResourceRequirements(requests={"amd.com/gpu": 1}))]))
+    lws = LeaderWorkerSet()
+    lws.metadata = ObjectMeta(name="bench-lws", namespace="default")
+    lws.spec = LeaderWorkerSetSpec(
+        replicas=1,
+        leader_worker_template=LeaderWorkerTemplate(size=size,
+                                                    worker_template=tmpl))
+    return lws
+
+
+def lws_available(cluster, updated_replicas: int | None = None):
+    cur = cluster.get_lws("default", "bench-lws")
+    if cur is None:
+        return None
+    cond = next((c for c in cur.status.conditions if c.type == "Available"),
+                None)
+    if cond is None or cond.status != "True":
+        return None
+    if updated_replicas is not None and \
+            cur.status.updated_replicas != updated_replicas:
+        return None
+    return cur
+
+
+def wait_until(fn, timeout: float, desc: str, poll: float = 0.002):
+    deadline = time.monotonic() + timeout
+    while time.monotonic() < deadline:
+        r = fn()
+        if r:
+            return r
+        time.sleep(poll)
+    raise TimeoutError(f"bench: timed out waiting for {desc}")
+
+
+class BenchConductor:
+    """Rank-0 driver: control plane + conductor + measurement loop."""
+
+    def __init__(self, args, world: int, device: str, control_group, sync):
+        from lws_amd.cluster.cluster import LwsCluster, make_nodes
+
+        self.args = args
+        self.world = world
+        self.device = device
+        self.runtime = CollectiveGroupRuntime(group_size=world)
+        self.conductor = Conductor(0, world, device, control_group, sync)
+        nodes = make_nodes(world, gpus_per_node=1, topology_per_node=False)
+        self.cluster = LwsCluster(
+            nodes=nodes, runtime_factory=lambda n: self.runtime).start()
+        self.env_checked = False
+
+    def _serve_builds_until(self, predicate, timeout: float, desc: str):
+        """Pump pending collective builds while waiting for a condition."""
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            r = predicate()
+            if r:
+                return r
+            rev = self.runtime.drain_pending_build(timeout=0.002)
+            if rev is not None:
+                spec = self.runtime.spec_from_pods(rev)
+                self.conductor.command({"op": "build", "spec": spec})
+                self.runtime.mark_revision_ready(rev)
+        raise TimeoutError(f"bench: timed out waiting for {desc}")
+
+    def _check_env_contract(self):
+        """Assert the RCCL/LWS env injection the framework promises."""
+        pods = self.cluster.store.list("Pod", "default")
+        assert pods, "no pods"
+        for pod in pods:
+            env = {e.name: e.value for e in pod.spec.containers[0].env}
+            widx = pod.metadata.labels[lwsapi.WORKER_INDEX_LABEL_KEY]
+            assert env[lwsapi.LWS_GROUP_SIZE] == str(self.world)
+            assert env[lwsapi.LWS_WORKER_INDEX] == widx
+            assert env[lwsapi.LWS_LEADER_ADDRESS].startswith("bench-lws-0.")
+            assert env["WORLD_SIZE"] == str(self.world)
+            assert env["NODE_RANK"] == widx
+        self.env_checked = True
+
+    def run_cycle(self, seed_extra: int) -> dict:
+        args = self.args
+        store = self.cluster.store
+        t0 = time.perf_counter()
+        lws = build_lws(args, self.world)
+        lws.spec.leader_worker_template.worker_template.metadata.annotations[
+            BENCH_SEED_ANNOTATION] = str(args.seed + seed_extra)
+        store.create(lws)
+        self._serve_builds_until(lambda: lws_available(self.cluster),
+                                 timeout=3600, desc="group Available")
+        t_ready = time.perf_counter() - t0
+        if not self.env_checked:
+            self._check_env_contract()
+        shard_info = dict(self.conductor.host.timings)
+
+        # rolling update: bump a template annotation -> new revision
+        cur = self.cluster.get_lws("default", "bench-lws")
+        old_rev_pods = {p.metadata.uid
+                        for p in store.list("Pod", "default")}
+        cur.spec.leader_worker_template.worker_template.metadata.annotations[
+            "bench.lws.amd.com/generation"] = str(seed_extra + 1)
+        t1 = time.perf_counter()
+        store.update(cur)
+
+        def rolled():
+            c = lws_available(self.cluster, updated_replicas=1)
+            if c is None:
+                return None
+            live = [p for p in store.list("Pod", "default")
+                    if p.metadata.deletion_timestamp is None]
+            if len(live) != self.world:
+                return None
+            if any(p.metadata.uid in old_rev_pods for p in live):
+                return None
+            return c
+        self._serve_builds_until(rolled, timeout=3600,
+                                 desc="rolling update complete")
+        t_rollout = time.perf_counter() - t1
+
+        # teardown
+        store.delete(lwsapi.KIND, "default", "bench-lws",
+                     propagation="Background")
+        wait_until(lambda: not store.list("Pod", "default"), 600,
+                   "pods drained")
+        self.conductor.command({"op": "teardown"})
+        return {"time_to_ready_s": t_ready, "rollout_s": t_rollout,
+                "shard": shard_info}
+
+    def decode_bench(self) -> dict:
+        args = self.args
+        spec = {"model": args.model, "kv_pages": args.kv_pages,
+                "seed": args.seed}
+        self.conductor.command({"op": "build", "spec": spec})
+        acks = self.conductor.command({
+            "op": "decode_bench", "batch": args.decode_batch,
+            "prompt_len": args.prompt_len, "steps": args.decode_steps})
+        self.conductor.command({"op": "teardown"})
+        return acks[0]
+
+    def shutdown(self):
+        self.conductor.command({"op": "exit"})
+        self.cluster.stop()
+
+
+def main() -> None:
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    if args.device is None:
+        args.device = "cuda" if torch.cuda.is_available() else "cpu"
+    is_gpu = args.device.startswith("cuda")
+
+    control_group = None
+    if world > 1:
+        backend = "nccl" if is_gpu else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+        if is_gpu:
+            torch.cuda.set_device(rank % torch.cuda.device_count())
+            args.device = f"cuda:{rank % torch.cuda.device_count()}"
+        control_group = dist.new_group(backend="gloo")
+        from lws_amd.parallel.tp import init_distributed
+        init_distributed(backend=backend, device=args.device)
+    elif is_gpu:
+        torch.cuda.set_device(0)
+        args.device = "cuda:0"
+
+    def sync():
+        if world > 1:
+            dist.barrier()
+        if is_gpu:
+            torch.cuda.synchronize()
+
+    if rank != 0:
+        # workers follow commands; block timing handled by execute_command
+        WorkerLoop(rank, world, args.device, control_group, sync).run()
+        return
+
+    # ---- rank 0 ----
+    bc = BenchConductor(args, world, args.device, control_group, sync)
+    try:
+        for w in range(args.warmup):
+            bc.run_cycle(seed_extra=1000 + w)
+
+        bc.conductor.command({"op": "block_begin"})
+        results = [bc.run_cycle(seed_extra=i) for i in range(args.steps)]
+        acks = bc.conductor.command({"op": "block_end"})
+        elapsed = max(a for a in acks if isinstance(a, float))
+
+        decode = None
+        if not args.skip_decode_bench:
+            decode = bc.decode_bench()
+
+        ready_ms = sorted(r["time_to_ready_s"] * 1000 for r in results)
+        rollout_ms = sorted(r["rollout_s"] * 1000 for r in results)
+        value = statistics.median(ready_ms)
+        shard = results[-1]["shard"]
+        out = {
+            "metric": METRIC,
+            "value": round(value, 2),
+            "unit": "ms",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1000 / args.steps, 2),
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.decode_batch,
+                "seq_len": args.prompt_len,
+                "parallelism": f"tp{world}",
+                "group": f"1-leader/{world - 1}-workers (size={world})",
+                "replicas": 1,
+                "time_to_ready_ms_p50": round(value, 2),
+                "time_to_ready_ms_all": [round(v, 2) for v in ready_ms],
+                "rollout_ms_p50": round(statistics.median(rollout_ms), 2),
+                "rollout_ms_all": [round(v, 2) for v in rollout_ms],
+                "shard_params": shard.get("params"),
+                "weights_s": round(shard.get("weights_s", 0), 3),
+                "warmup_s": round(shard.get("warmup_s", 0), 3),
+                "decode_tokens_per_s": (round(decode["tokens_per_s"], 1)
+                                        if decode else None),
+                "orchestrator": "lws_amd in-process control plane",
+            },
+        }
+        print(json.dumps(out), flush=True)
+    finally:
+        bc.shutdown()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,213 @@
+"""Collective group runtime: bridges the control plane to engine shards.
+
+In an lws_amd deployment one group = N pods = N processes, one per MI355X
+GPU, rendezvoused over RCCL via the env the pod webhook injects.  In the
+single-node benchmark/test harness, those N processes are the launcher's
+ranks: rank 0 hosts the control plane (store/controllers/scheduler/agents)
+plus shard 0; ranks 1..N-1 run a command loop and host shards 1..N-1.
+
+The conductor (rank 0) aggregates pod lifecycle events from the node
+agents; when every pod of a revision has started, it broadcasts a BUILD
+command over the gloo control group.  All ranks then build their engine
+shard together (weight materialization + KV pool + warmup collectives over
+RCCL), ack, and rank 0 marks the pods Ready — which is exactly the
+"replica-group time-to-ready" path the BASELINE metric measures.
+"""
+from __future__ import annotations
+
+import os
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..api import leaderworkerset as lwsapi
+from ..cluster.node import NodeAgent, PodRuntime
+from ..serving.engine import Engine, EngineConfig
+
+BENCH_MODEL_ANNOTATION = "bench.lws.amd.com/model"
+BENCH_KV_PAGES_ANNOTATION = "bench.lws.amd.com/kv-pages"
+BENCH_SEED_ANNOTATION = "bench.lws.amd.com/seed"
+
+
+@dataclass
+class ShardHost:
+    """Engine shard owner for one rank (any rank, incl. the conductor)."""
+
+    rank: int
+    world: int
+    device: str
+    engine: Optional[Engine] = None
+    timings: dict = field(default_factory=dict)
+
+    def build(self, spec: dict) -> dict:
+        self.teardown()
+        cfg = EngineConfig(model=spec["model"],
+                           kv_pages=int(spec.get("kv_pages", 128)),
+                           seed=int(spec.get("seed", 0)),
+                           device=self.device, tp_rank=self.rank,
+                           tp_world=self.world)
+        self.engine = Engine(cfg)
+        info = self.engine.load()
+        self.timings = info
+        return info
+
+    def teardown(self) -> None:
+        if self.engine is not None:
+            self.engine.unload()
+            self.engine = None
+
+    def decode_bench(self, batch: int, prompt_len: int, steps: int) -> dict:
+        """Timed decode loop (whole group enters together)."""
+        assert self.engine is not None
+        eng = self.engine
+        prompts = [[(i * 7 + j) % eng.model_cfg.vocab_size
+                    for j in range(prompt_len)] for i in range(batch)]
+        sids = [eng.add_request(p) for p in prompts]
+        eng.step()  # prefill
+        # warmup decodes
+        for _ in range(2):
+            eng.step()
+        if self.device.startswith("cuda"):
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            eng.step()
+        if self.device.startswith("cuda"):
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        for s in sids:
+            eng.finish(s)
+        return {"decode_steps": steps, "batch": batch, "seconds": dt,
+                "tokens_per_s": batch * steps / dt}
+
+
+def execute_command(cmd: dict, host: ShardHost, sync, t_block: dict) -> Any:
+    """Shared command execution — runs in the SAME order on every rank so
+    collectives (default-group barriers inside sync, engine build/decode
+    all-reduces) always line up."""
+    op = cmd["op"]
+    if op == "build":
+        return host.build(cmd["spec"])
+    if op == "teardown":
+        host.teardown()
+        return None
+    if op == "decode_bench":
+        return host.decode_bench(cmd["batch"], cmd["prompt_len"], cmd["steps"])
+    if op == "block_begin":
+        sync()
+        t_block["t0"] = time.perf_counter()
+        return None
+    if op == "block_end":
+        sync()
+        return time.perf_counter() - t_block.get("t0", time.perf_counter())
+    if op == "exit":
+        host.teardown()
+        return None
+    return None
+
+
+class WorkerLoop:
+    """Ranks 1..N-1: execute commands broadcast by the conductor."""
+
+    def __init__(self, rank: int, world: int, device: str, control_group,
+                 sync=lambda: None):
+        self.host = ShardHost(rank, world, device)
+        self.group = control_group
+        self.sync = sync
+        self.t_block: dict = {}
+
+    def run(self) -> None:
+        while True:
+            buf = [None]
+            dist.broadcast_object_list(buf, src=0, group=self.group)
+            cmd = buf[0]
+            result = execute_command(cmd, self.host, self.sync, self.t_block)
+            acks: list = [None] * dist.get_world_size(self.group)
+            dist.all_gather_object(acks, result, group=self.group)
+            if cmd["op"] == "exit":
+                return
+
+
+class Conductor:
+    """Rank 0: drives commands and owns shard 0."""
+
+    def __init__(self, rank: int, world: int, device: str, control_group,
+                 sync=lambda: None):
+        self.host = ShardHost(rank, world, device)
+        self.group = control_group
+        self.world = world
+        self.sync = sync
+        self.t_block: dict = {}
+
+    def command(self, cmd: dict) -> list:
+        if self.world > 1:
+            dist.broadcast_object_list([cmd], src=0, group=self.group)
+        result = execute_command(cmd, self.host, self.sync, self.t_block)
+        if self.world > 1:
+            acks: list = [None] * dist.get_world_size(self.group)
+            dist.all_gather_object(acks, result, group=self.group)
+            return acks
+        return [result]
+
+
+class CollectiveGroupRuntime(PodRuntime):
+    """PodRuntime for rank 0's node agents: per-pod start/stop events are
+    aggregated per revision; the bench main loop executes pending builds
+    via the Conductor and then marks the revision's pods ready."""
+
+    def __init__(self, group_size: int):
+        self.group_size = group_size
+        self._lock = threading.Lock()
+        # revision -> {pod_name: (pod, agent)}
+        self.started: dict[str, dict[str, tuple]] = {}
+        self.built_revision: Optional[str] = None
+        self.events = queue.Queue()
+
+    def start(self, pod, agent: NodeAgent) -> None:
+        rev = (pod.metadata.labels or {}).get(lwsapi.REVISION_KEY, "")
+        with self._lock:
+            pods = self.started.setdefault(rev, {})
+            if pod.metadata.name in pods:
+                return
+            pods[pod.metadata.name] = (pod, agent)
+            if len(pods) == self.group_size:
+                self.events.put(("build", rev))
+
+    def stop(self, pod, agent: NodeAgent) -> None:
+        rev = (pod.metadata.labels or {}).get(lwsapi.REVISION_KEY, "")
+        with self._lock:
+            pods = self.started.get(rev, {})
+            pods.pop(pod.metadata.name, None)
+        agent.finish_pod_teardown(pod)
+
+    def spec_from_pods(self, rev: str) -> dict:
+        with self._lock:
+            pods = self.started.get(rev, {})
+            pod, _ = next(iter(pods.values()))
+        ann = pod.metadata.annotations or {}
+        return {
+            "model": ann.get(BENCH_MODEL_ANNOTATION, "llama-tiny"),
+            "kv_pages": ann.get(BENCH_KV_PAGES_ANNOTATION, "128"),
+            "seed": ann.get(BENCH_SEED_ANNOTATION, "0"),
+            "revision": rev,
+        }
+
+    def mark_revision_ready(self, rev: str) -> None:
+        with self._lock:
+            pods = list(self.started.get(rev, {}).values())
+        for pod, agent in pods:
+            agent.mark_pod_ready(pod)
+        self.built_revision = rev
+
+    def drain_pending_build(self, timeout: float) -> Optional[str]:
+        """Wait for the next complete-revision event."""
+        try:
+            kind, rev = self.events.get(timeout=timeout)
+            return rev
+        except queue.Empty:
+            return None

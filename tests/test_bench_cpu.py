@@ -1,0 +1,76 @@
+"""CPU dry-runs of bench.py — the same path the driver runs on MI355X,
+with llama-tiny on the fp32 reference ops and gloo collectives."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_bench(env_extra, args):
+    env = dict(os.environ)
+    env.update(env_extra)
+    cmd = [sys.executable, os.path.join(REPO, "bench.py"), "--device", "cpu",
+           "--model", "llama-tiny", "--kv-pages", "64", "--decode-batch", "4",
+           "--prompt-len", "16", "--decode-steps", "4"] + args
+    return subprocess.run(cmd, capture_output=True, text=True, env=env,
+                          cwd=REPO, timeout=600)
+
+
+def _parse_json_line(stdout):
+    for line in stdout.splitlines():
+        line = line.strip()
+        if line.startswith("{"):
+            return json.loads(line)
+    raise AssertionError(f"no JSON line in output:\n{stdout}")
+
+
+def test_bench_single_process():
+    r = _run_bench({}, ["--gpus", "1", "--steps", "2", "--warmup", "1"])
+    assert r.returncode == 0, r.stderr[-4000:]
+    out = _parse_json_line(r.stdout)
+    assert out["n_gpus"] == 1
+    assert out["steps"] == 2
+    assert out["higher_is_better"] is False
+    assert out["value"] > 0
+    assert out["config"]["rollout_ms_p50"] > 0
+    assert out["config"]["decode_tokens_per_s"] > 0
+    assert out["config"]["parallelism"] == "tp1"
+
+
+def test_bench_two_rank_gloo():
+    """Two ranks over gloo — the distributed path the driver exercises with
+    torch.distributed.run on the GPU box."""
+    import random
+    port = random.randint(20000, 40000)
+    procs = []
+    for rank in range(2):
+        env = {"RANK": str(rank), "WORLD_SIZE": "2",
+               "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+               "LOCAL_RANK": str(rank)}
+        p_env = dict(os.environ)
+        p_env.update(env)
+        cmd = [sys.executable, os.path.join(REPO, "bench.py"),
+               "--device", "cpu", "--model", "llama-tiny", "--kv-pages", "64",
+               "--decode-batch", "4", "--prompt-len", "16",
+               "--decode-steps", "4", "--gpus", "2", "--steps", "1",
+               "--warmup", "0"]
+        procs.append(subprocess.Popen(cmd, stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True,
+                                      env=p_env, cwd=REPO))
+    outs = []
+    for p in procs:
+        try:
+            stdout, stderr = p.communicate(timeout=600)
+        except subprocess.TimeoutExpired:
+            p.kill()
+            raise
+        assert p.returncode == 0, stderr[-4000:]
+        outs.append(stdout)
+    out = _parse_json_line(outs[0])
+    assert out["n_gpus"] == 2
+    assert out["config"]["parallelism"] == "tp2"
+    assert out["value"] > 0
