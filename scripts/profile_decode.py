@@ -1,0 +1,60 @@
+#!/usr/bin/env python3
+"""Clean decode-loop profiling target for rocprofv3 (no control plane)."""
+import argparse
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import torch  # noqa: E402
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--batch", type=int, default=32)
+    p.add_argument("--prompt-len", type=int, default=128)
+    p.add_argument("--steps", type=int, default=16)
+    p.add_argument("--kv-pages", type=int, default=512)
+    p.add_argument("--mode", default="decode", choices=["decode", "prefill"])
+    args = p.parse_args()
+
+    from lws_amd.serving.engine import Engine, EngineConfig
+
+    eng = Engine(EngineConfig(model=args.model, kv_pages=args.kv_pages,
+                              device="cuda"))
+    info = eng.load()
+    print("load:", info, flush=True)
+    prompts = [[(i * 7 + j) % eng.model_cfg.vocab_size
+                for j in range(args.prompt_len)] for i in range(args.batch)]
+    if args.mode == "decode":
+        sids = [eng.add_request(pr) for pr in prompts]
+        eng.step()   # prefill
+        eng.step()   # warm decode
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            eng.step()
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        print(f"decode: {args.steps} steps, batch {args.batch}: "
+              f"{args.batch * args.steps / dt:.1f} tok/s, "
+              f"{dt / args.steps * 1000:.2f} ms/step", flush=True)
+    else:
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for i in range(args.steps):
+            sids = [eng.add_request(pr) for pr in prompts]
+            eng.step()
+            for s in sids:
+                eng.finish(s)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        toks = args.batch * args.prompt_len * args.steps
+        print(f"prefill: {toks / dt:.0f} tok/s, "
+              f"{dt / args.steps * 1000:.1f} ms/step", flush=True)
+
+
+if __name__ == "__main__":
+    main()
