@@ -228,6 +228,30 @@ class LlamaForCausalLM:
             k_cache[s // page, :, s % page] = k[t]
             v_cache[s // page, :, s % page] = v[t]
 
+    def _qkv_views(self, qkv: torch.Tensor, layer, positions):
+        """Split the fused qkv GEMM output into q/k/v [T, H, D] and apply
+        rope + return views.  On GPU these are zero-copy strided views into
+        the qkv buffer (the HIP kernels take row strides); on CPU (reference
+        path) contiguous copies."""
+        d = self.cfg.head_dim
+        if self._ops.is_gpu:
+            q = qkv.narrow(-1, 0, layer.q_slice).unflatten(-1, (layer.hq, d))
+            k = qkv.narrow(-1, layer.q_slice,
+                           layer.kv_slice).unflatten(-1, (layer.hkv, d))
+            v = qkv.narrow(-1, layer.q_slice + layer.kv_slice,
+                           layer.kv_slice).unflatten(-1, (layer.hkv, d))
+            self._ops.ops.rope(q, k, self.rope_table, positions,
+                               layer.hq, layer.hkv)
+            return q, k, v
+        q, k, v = qkv.split([layer.q_slice, layer.kv_slice, layer.kv_slice],
+                            dim=-1)
+        q, k = self._ops.ref.rope_ref(q.contiguous(), k.contiguous(),
+                                      self.rope_table, positions,
+                                      layer.hq, layer.hkv)
+        T = qkv.size(0)
+        return (q.view(T, layer.hq, d), k.view(T, layer.hkv, d),
+                v.contiguous().view(T, layer.hkv, d))
+
     # -- forward --------------------------------------------------------
     def forward_prefill(self, batch: PrefillBatch, kv_caches) -> torch.Tensor:
         """Returns hidden states of the LAST token of each sequence [B, H]."""
@@ -242,15 +266,9 @@ class LlamaForCausalLM:
                 h, residual = self._fused_add_rmsnorm(x, residual,
                                                       layer.input_norm)
             qkv = layer.qkv(h)
-            q, k, v = qkv.split([layer.q_slice, layer.kv_slice,
-                                 layer.kv_slice], dim=-1)
-            q = q.contiguous()
-            k = k.contiguous()
-            q, k = self._rope(q, k, batch.positions, layer.hq, layer.hkv)
+            q, k, v = self._qkv_views(qkv, layer, batch.positions)
             k_cache, v_cache = kv_caches[li]
-            self._write_cache(k.view(-1, layer.hkv, cfg.head_dim),
-                              v.contiguous().view(-1, layer.hkv, cfg.head_dim),
-                              k_cache, v_cache, batch.slot_mapping)
+            self._write_cache(k, v, k_cache, v_cache, batch.slot_mapping)
             attn = self._prefill_attention(q, k, v, batch, layer)
             o = layer.o(attn)
             x = all_reduce(o)
@@ -265,9 +283,9 @@ class LlamaForCausalLM:
         return h[last]
 
     def _prefill_attention(self, q, k, v, batch: PrefillBatch, layer):
-        """Per-sequence causal attention (library-GEMM composition in bf16
-        with fp32 softmax; the decode path is the hot loop and uses the HIP
-        paged-attention kernel)."""
+        """Per-sequence causal attention over [T, H, D] views (library-GEMM
+        composition in bf16 with fp32 softmax; the decode path is the hot
+        loop and uses the HIP paged-attention kernel)."""
         cfg = self.cfg
         d = cfg.head_dim
         T = q.size(0)
@@ -276,9 +294,9 @@ class LlamaForCausalLM:
         for i in range(len(batch.seq_starts) - 1):
             s0, s1 = batch.seq_starts[i], batch.seq_starts[i + 1]
             S = s1 - s0
-            qs = q[s0:s1].view(S, layer.hq, d).transpose(0, 1)      # [Hq,S,D]
-            ks = k[s0:s1].view(S, layer.hkv, d).transpose(0, 1)
-            vs = v[s0:s1].view(S, layer.hkv, d).transpose(0, 1)
+            qs = q[s0:s1].transpose(0, 1)      # [Hq, S, D]
+            ks = k[s0:s1].transpose(0, 1)
+            vs = v[s0:s1].transpose(0, 1)
             ks = ks.repeat_interleave(G, dim=0)
             vs = vs.repeat_interleave(G, dim=0)
             scores = (qs @ ks.transpose(-1, -2)).float() * self.scale
@@ -302,25 +320,18 @@ class LlamaForCausalLM:
                 h, residual = self._fused_add_rmsnorm(x, residual,
                                                       layer.input_norm)
             qkv = layer.qkv(h)
-            q, k, v = qkv.split([layer.q_slice, layer.kv_slice,
-                                 layer.kv_slice], dim=-1)
-            q = q.contiguous()
-            k = k.contiguous()
-            q, k = self._rope(q, k, batch.positions, layer.hq, layer.hkv)
+            q, k, v = self._qkv_views(qkv, layer, batch.positions)
             k_cache, v_cache = kv_caches[li]
-            self._write_cache(k.view(-1, layer.hkv, cfg.head_dim),
-                              v.contiguous().view(-1, layer.hkv, cfg.head_dim),
-                              k_cache, v_cache, batch.slot_mapping)
+            self._write_cache(k, v, k_cache, v_cache, batch.slot_mapping)
             B = q.size(0)
-            qh = q.view(B, layer.hq, cfg.head_dim)
             if self._ops.is_gpu:
                 attn = self._ops.ops.paged_attention_decode(
-                    qh, k_cache, v_cache, batch.block_tables, batch.seq_lens,
+                    q, k_cache, v_cache, batch.block_tables, batch.seq_lens,
                     self.scale)
             else:
                 attn = self._ops.ref.paged_attention_decode_ref(
-                    qh, k_cache, v_cache, batch.block_tables, batch.seq_lens,
-                    self.scale)
+                    q.contiguous(), k_cache, v_cache, batch.block_tables,
+                    batch.seq_lens, self.scale)
             o = layer.o(attn.view(B, layer.hq * cfg.head_dim))
             x = all_reduce(o)
             h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)

@@ -110,7 +110,7 @@ def build_rope_table(max_pos: int, head_dim: int, theta: float = 10000.0,
 def paged_attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
                            v_cache: torch.Tensor, block_tables: torch.Tensor,
                            seq_lens: torch.Tensor, scale: float,
-                           chunk_keys: int = 512,
+                           chunk_keys: Optional[int] = None,
                            workspace: Optional[tuple] = None,
                            out: Optional[torch.Tensor] = None) -> torch.Tensor:
     lib = require_native()
@@ -118,6 +118,13 @@ def paged_attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
     Hkv = k_cache.size(1)
     G = Hq // Hkv
     max_len = int(block_tables.size(1)) * int(k_cache.size(2))
+    if chunk_keys is None:
+        # flash-decoding split: size chunks so the launch approaches ~1024
+        # workgroups (4 waves each) and 256 CUs stay busy even at B*Hkv=1
+        target_chunks = max(1, 1024 // max(1, B * Hkv))
+        target_chunks = min(target_chunks, (max_len + 63) // 64)
+        chunk_keys = -(-max_len // target_chunks)
+        chunk_keys = ((chunk_keys + 15) // 16) * 16
     num_chunks = max(1, (max_len + chunk_keys - 1) // chunk_keys)
     if workspace is None:
         ws_acc = torch.empty((B, Hkv, num_chunks, G, D), dtype=torch.float32,

@@ -1,67 +1,68 @@
 // Paged-attention decode for gfx950 (MI355X), GQA-aware, flash-decoding
-// sequence split.
+// sequence split.  v2 design notes:
 //
-// Layout (one KV read amortized over the whole GQA group — the right CDNA4
-// decode design for Llama-70B TP shards where H_kv per GPU is small):
-//   grid = (num_seqs, H_kv, seq_chunks)  x  block 256 threads (4 waves)
-//   Each workgroup processes CHUNK keys of one sequence for one kv head and
-//   ALL G q-heads of that kv head's GQA group (G <= 16).
-//   Thread (key, slice) layout: lane = key_local*16 + slice; each lane loads
-//   16 B (8 bf16) of the key row -> fully-coalesced 256 B per key row, 16
-//   keys in flight per pass over the workgroup (64 keys per chunk pass with
-//   4 waves).  Scores reduce within 16-lane groups via shfl; online softmax
-//   per chunk; V accumulated in registers per (key, slice) and reduced
-//   across keys at the end.
-//   Chunk partials (m, l, acc) land in an fp32 workspace; a second kernel
-//   reduces chunks (flash-decoding), so 256 CUs stay busy even at B=1.
+//  - grid = (num_seqs, H_kv, seq_chunks) x 256 threads (4 waves); the host
+//    wrapper sizes seq_chunks so the launch fills the 256-CU chip even at
+//    small B*H_kv (one kv head per GPU at TP8).
+//  - Each WAVE owns an independent interleaved key stream with its own
+//    online-softmax state: NO __syncthreads in the key loop (v1 was
+//    latency-bound on two block syncs per 16-key pass).
+//  - Wave lane layout: lane = kgrp*16 + slice; 4 keys in flight per wave
+//    pass, each key read by 16 lanes x 16 B = 256 B fully-coalesced rows.
+//  - One KV read is amortized over the whole GQA group (G q-heads).
+//  - Wave partials (m, l, acc) merge once at the end through LDS, then
+//    chunk partials merge in a tiny reduce kernel (flash-decoding).
+//  - q may be strided (read straight out of the fused qkv GEMM buffer --
+//    no .contiguous() copies on the decode path).
 //
-// KV cache layout: [num_pages, H_kv, page_size, D] bf16, D = 128, page 16.
+// KV cache layout: [num_pages, H_kv, page_size, D] bf16, D = 128.
 #include "common.h"
 
 #define PA_HEAD_DIM 128
 #define PA_SLICES 16                // 16 lanes x 8 bf16 = 128 elements
-#define PA_KEYS_PER_PASS 16         // 256 threads / 16 slices
+#define PA_NWAVES 4
 #define PA_MAX_GQA 16
 
 __global__ __launch_bounds__(256)
 void paged_attention_chunk_kernel(
     float* __restrict__ ws_acc,      // [B, Hkv, chunks, G, 128]
     float* __restrict__ ws_ml,       // [B, Hkv, chunks, G, 2]  (m, l)
-    const ushort* __restrict__ q,    // [B, Hq, 128]
+    const ushort* __restrict__ q,    // [B, Hq, 128] (row stride q_stride)
     const ushort* __restrict__ k_cache,  // [pages, Hkv, page, 128]
     const ushort* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [B, max_pages]
     const int* __restrict__ seq_lens,      // [B]
     float scale, int G, int Hkv, int page_size, int max_pages,
-    int chunk_keys, int num_chunks) {
+    int chunk_keys, int num_chunks, long long q_stride) {
   const int b = blockIdx.x;
   const int hkv = blockIdx.y;
   const int chunk = blockIdx.z;
   const int seq_len = seq_lens[b];
   const int kstart = chunk * chunk_keys;
-  if (kstart >= seq_len && chunk > 0) {
-    // out-of-range chunk: mark empty partial
-    if (threadIdx.x < G) {
-      const long long mlbase =
-          ((((long long)b * Hkv + hkv) * num_chunks + chunk) * G + threadIdx.x) * 2;
-      ws_ml[mlbase] = -INFINITY;
-      ws_ml[mlbase + 1] = 0.0f;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int slice = lane % PA_SLICES;       // 8-elem slice of the head dim
+  const int kgrp = lane / PA_SLICES;        // key group within the wave
+
+  if (kstart >= seq_len) {
+    if (chunk > 0 || seq_len > 0) {
+      if (threadIdx.x < G) {
+        const long long mlbase =
+            ((((long long)b * Hkv + hkv) * num_chunks + chunk) * G
+             + threadIdx.x) * 2;
+        ws_ml[mlbase] = -INFINITY;
+        ws_ml[mlbase + 1] = 0.0f;
+      }
+      return;
     }
-    return;
   }
   const int kend = min(kstart + chunk_keys, seq_len);
 
-  const int lane = threadIdx.x % WAVE_SIZE;
-  const int wave = threadIdx.x / WAVE_SIZE;
-  const int slice = threadIdx.x % PA_SLICES;        // which 8-elem slice
-  const int key_local = threadIdx.x / PA_SLICES;    // 0..15
-
   // q for the whole GQA group staged in LDS: [G][128] fp32 (pre-scaled)
   __shared__ float q_lds[PA_MAX_GQA][PA_HEAD_DIM];
-  __shared__ float score_lds[PA_MAX_GQA][PA_KEYS_PER_PASS];
-
   for (int g = 0; g < G; ++g) {
-    const ushort* qrow = q + (((long long)b * Hkv + hkv) * G + g) * PA_HEAD_DIM;
+    const ushort* qrow = q + (long long)b * q_stride
+        + ((long long)hkv * G + g) * PA_HEAD_DIM;
     for (int i = threadIdx.x * 8; i < PA_HEAD_DIM; i += blockDim.x * 8) {
       bf16x8 v;
       v.u = *reinterpret_cast<const uint4*>(qrow + i);
@@ -71,9 +72,8 @@ void paged_attention_chunk_kernel(
   }
   __syncthreads();
 
-  // online-softmax state per q-head (uniform across the workgroup)
+  // per-wave online-softmax state + per-thread V accumulator
   float m_run[PA_MAX_GQA], l_run[PA_MAX_GQA];
-  // per-thread V accumulator: this thread's (key stream, slice) partial
   float acc[PA_MAX_GQA][8];
   for (int g = 0; g < G; ++g) {
     m_run[g] = -INFINITY;
@@ -82,86 +82,75 @@ void paged_attention_chunk_kernel(
     for (int j = 0; j < 8; ++j) acc[g][j] = 0.0f;
   }
 
-  for (int k0 = kstart; k0 < kend; k0 += PA_KEYS_PER_PASS) {
-    const int key = k0 + key_local;
+  const long long kv_head_base = (long long)hkv * page_size * PA_HEAD_DIM;
+  const long long kv_page_stride = (long long)Hkv * page_size * PA_HEAD_DIM;
+
+  // wave-interleaved key streams: wave w takes keys [kstart+w*4+kgrp],
+  // stepping 16 keys per workgroup pass — no block-level sync inside.
+  for (int k0 = kstart + wave * 4; k0 < kend; k0 += 4 * PA_NWAVES) {
+    const int key = k0 + kgrp;
     const bool valid = key < kend;
-    // locate the key's page
-    bf16x8 kv;
-    kv.u = make_uint4(0, 0, 0, 0);
-    const ushort* vrow = nullptr;
+    float kf[8];
+    long long row = 0;
     if (valid) {
       const int page = block_tables[(long long)b * max_pages + key / page_size];
-      const int off = key % page_size;
-      const ushort* krow = k_cache +
-          ((((long long)page * Hkv + hkv) * page_size + off) * PA_HEAD_DIM);
-      vrow = v_cache +
-          ((((long long)page * Hkv + hkv) * page_size + off) * PA_HEAD_DIM);
-      kv.u = *reinterpret_cast<const uint4*>(krow + slice * 8);
-    }
-    float kf[8];
+      row = (long long)page * kv_page_stride + kv_head_base
+          + (long long)(key % page_size) * PA_HEAD_DIM;
+      bf16x8 kv8;
+      kv8.u = *reinterpret_cast<const uint4*>(k_cache + row + slice * 8);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv.h[j]);
-
-    // scores for all G heads: partial dot over this lane's 8 elems,
-    // reduced across the 16 lanes of the key group
+      for (int j = 0; j < 8; ++j) kf[j] = bf16_to_f32(kv8.h[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) kf[j] = 0.0f;
+    }
+    // scores for all G heads (dot over this lane's slice, reduced across
+    // the 16 lanes of the key group), then per-wave online softmax
+    float vf[8];
+    bool v_loaded = false;
     for (int g = 0; g < G; ++g) {
       float p = 0.0f;
 #pragma unroll
       for (int j = 0; j < 8; ++j) p += kf[j] * q_lds[g][slice * 8 + j];
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) p += __shfl_xor(p, off, WAVE_SIZE);
-      // lane slice==0 of each key group holds the full dot
-      if (slice == 0) score_lds[g][key_local] = valid ? p : -INFINITY;
-    }
-    __syncthreads();
-
-    // softmax update (uniform): each thread reads the 16 scores
-    float pexp[PA_MAX_GQA];
-    for (int g = 0; g < G; ++g) {
-      float m_new = m_run[g];
-#pragma unroll
-      for (int i = 0; i < PA_KEYS_PER_PASS; ++i)
-        m_new = fmaxf(m_new, score_lds[g][i]);
-      float rescale = (m_run[g] == -INFINITY) ? 0.0f : __expf(m_run[g] - m_new);
-      float s = score_lds[g][key_local];
-      float e = (valid && s != -INFINITY) ? __expf(s - m_new) : 0.0f;
-      pexp[g] = e;
-      float lsum = 0.0f;
-#pragma unroll
-      for (int i = 0; i < PA_KEYS_PER_PASS; ++i) {
-        float si = score_lds[g][i];
-        lsum += (si == -INFINITY) ? 0.0f : __expf(si - m_new);
-      }
-      l_run[g] = l_run[g] * rescale + lsum;
+      float s_local = valid ? p : -INFINITY;
+      // wave-wide max over the 4 key groups
+      float m4 = s_local;
+      m4 = fmaxf(m4, __shfl_xor(m4, 16, WAVE_SIZE));
+      m4 = fmaxf(m4, __shfl_xor(m4, 32, WAVE_SIZE));
+      const float m_new = fmaxf(m_run[g], m4);
+      if (m_new == -INFINITY) continue;  // nothing valid yet for this wave
+      const float rescale =
+          (m_run[g] == -INFINITY) ? 0.0f : __expf(m_run[g] - m_new);
+      const float e = (s_local == -INFINITY) ? 0.0f : __expf(s_local - m_new);
+      float l4 = e;
+      l4 += __shfl_xor(l4, 16, WAVE_SIZE);
+      l4 += __shfl_xor(l4, 32, WAVE_SIZE);
+      l_run[g] = l_run[g] * rescale + l4;
       m_run[g] = m_new;
+      if (!v_loaded && valid) {
+        bf16x8 vv;
+        vv.u = *reinterpret_cast<const uint4*>(v_cache + row + slice * 8);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc[g][j] *= rescale;
-    }
-    __syncthreads();
-
-    // V accumulate: this thread's key contributes p * v[slice]
-    if (valid) {
-      bf16x8 vv;
-      vv.u = *reinterpret_cast<const uint4*>(vrow + slice * 8);
-      float vf[8];
+        for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.h[j]);
+        v_loaded = true;
+      }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.h[j]);
-      for (int g = 0; g < G; ++g) {
-        const float p = pexp[g];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) acc[g][j] += p * vf[j];
+      for (int j = 0; j < 8; ++j) {
+        acc[g][j] = acc[g][j] * rescale + (valid ? e * vf[j] : 0.0f);
       }
     }
   }
 
-  // reduce acc over keys.  Within a wave, lanes with the same slice and
-  // different key_local differ in lane bits 4,5 -> shfl_xor(16)+shfl_xor(32)
-  // sums the wave's 4 key streams; every lane then holds its wave's slice
-  // partial.  Cross-wave partials reduce through LDS.
+  // merge the 4 waves' (m, l, acc) through LDS with softmax rescaling.
+  __shared__ float mw[PA_MAX_GQA][PA_NWAVES];
+  __shared__ float lw[PA_MAX_GQA][PA_NWAVES];
+  __shared__ float aw[PA_NWAVES][PA_SLICES][8];
   const long long wsbase =
       (((long long)b * Hkv + hkv) * num_chunks + chunk) * G * PA_HEAD_DIM;
-  __shared__ float xwave[4][PA_SLICES][8];
   for (int g = 0; g < G; ++g) {
+    // within-wave: sum acc over the 4 key groups (lane bits 4,5)
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float v = acc[g][j];
@@ -169,25 +158,36 @@ void paged_attention_chunk_kernel(
       v += __shfl_xor(v, 32, WAVE_SIZE);
       acc[g][j] = v;
     }
+    if (lane == 0) {
+      mw[g][wave] = m_run[g];
+      lw[g][wave] = l_run[g];
+    }
     if (lane < PA_SLICES) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xwave[wave][slice][j] = acc[g][j];
+      for (int j = 0; j < 8; ++j) aw[wave][slice][j] = acc[g][j];
     }
     __syncthreads();
-    // threads 0..127: (slice, j) sums over 4 waves
     if (threadIdx.x < PA_SLICES * 8) {
       const int s = threadIdx.x / 8;
       const int j = threadIdx.x % 8;
-      float v = xwave[0][s][j] + xwave[1][s][j] + xwave[2][s][j] + xwave[3][s][j];
-      ws_acc[wsbase + (long long)g * PA_HEAD_DIM + s * 8 + j] = v;
+      float m_g = fmaxf(fmaxf(mw[g][0], mw[g][1]), fmaxf(mw[g][2], mw[g][3]));
+      float accv = 0.0f, lv = 0.0f;
+#pragma unroll
+      for (int w = 0; w < PA_NWAVES; ++w) {
+        const float mwv = mw[g][w];
+        const float wgt = (mwv == -INFINITY) ? 0.0f : __expf(mwv - m_g);
+        accv += aw[w][s][j] * wgt;
+        lv += lw[g][w] * wgt;
+      }
+      ws_acc[wsbase + (long long)g * PA_HEAD_DIM + s * 8 + j] = accv;
+      if (s == 0 && j == 0) {
+        const long long mlbase =
+            ((((long long)b * Hkv + hkv) * num_chunks + chunk) * G + g) * 2;
+        ws_ml[mlbase] = m_g;
+        ws_ml[mlbase + 1] = lv;
+      }
     }
     __syncthreads();
-  }
-  if (threadIdx.x < G) {
-    const long long mlbase =
-        ((((long long)b * Hkv + hkv) * num_chunks + chunk) * G + threadIdx.x) * 2;
-    ws_ml[mlbase] = m_run[threadIdx.x];
-    ws_ml[mlbase + 1] = l_run[threadIdx.x];
   }
 }
 
@@ -208,7 +208,6 @@ void paged_attention_reduce_kernel(
   const int Hq = Hkv * G;
   const int used = min(num_chunks,
                        (seq_lens[b] + chunk_keys - 1) / chunk_keys);
-  // global max
   float m_glob = -INFINITY;
   for (int c = 0; c < used; ++c) {
     const long long mlbase =
@@ -236,12 +235,12 @@ void paged_attention_reduce_kernel(
 
 // ---------------------------------------------------------------------------
 // reshape_and_cache: scatter new k/v token rows into the paged cache.
-// k,v: [T, Hkv, 128]; slot_mapping: [T] (page*page_size + offset)
+// k,v: [T, Hkv, 128] with row stride (elements); slot = page*page_size+off
 __global__ void reshape_and_cache_kernel(
     const ushort* __restrict__ k, const ushort* __restrict__ v,
     ushort* __restrict__ k_cache, ushort* __restrict__ v_cache,
     const long long* __restrict__ slot_mapping,
-    int T, int Hkv, int page_size) {
+    int T, int Hkv, int page_size, long long k_stride, long long v_stride) {
   const long long total = (long long)T * Hkv * (PA_HEAD_DIM / 8);
   long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long stride = (long long)gridDim.x * blockDim.x;
@@ -254,13 +253,12 @@ __global__ void reshape_and_cache_kernel(
     if (slot < 0) continue;
     const long long page = slot / page_size;
     const long long off = slot % page_size;
-    const long long src = (((long long)t * Hkv + h) * PA_HEAD_DIM) + s8 * 8;
     const long long dst =
         ((((long long)page * Hkv + h) * page_size + off) * PA_HEAD_DIM) + s8 * 8;
-    *reinterpret_cast<uint4*>(k_cache + dst) =
-        *reinterpret_cast<const uint4*>(k + src);
-    *reinterpret_cast<uint4*>(v_cache + dst) =
-        *reinterpret_cast<const uint4*>(v + src);
+    *reinterpret_cast<uint4*>(k_cache + dst) = *reinterpret_cast<const uint4*>(
+        k + (long long)t * k_stride + (long long)h * PA_HEAD_DIM + s8 * 8);
+    *reinterpret_cast<uint4*>(v_cache + dst) = *reinterpret_cast<const uint4*>(
+        v + (long long)t * v_stride + (long long)h * PA_HEAD_DIM + s8 * 8);
   }
 }
 
@@ -274,8 +272,10 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                             torch::Tensor ws_acc, torch::Tensor ws_ml,
                             double scale, long long chunk_keys) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous() &&
-              k_cache.is_contiguous() && v_cache.is_contiguous());
+  TORCH_CHECK(out.is_contiguous() && k_cache.is_contiguous() &&
+              v_cache.is_contiguous());
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == PA_HEAD_DIM,
+              "q heads must be contiguous (row stride may differ)");
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
   const int B = q.size(0);
@@ -285,7 +285,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
   const int page_size = k_cache.size(2);
   const int G = Hq / Hkv;
   TORCH_CHECK(Hq % Hkv == 0 && G <= PA_MAX_GQA, "GQA group must be <= 16");
-  TORCH_CHECK(chunk_keys % PA_KEYS_PER_PASS == 0);
+  TORCH_CHECK(chunk_keys % PA_SLICES == 0);
   const int max_pages = block_tables.size(1);
   const int num_chunks = ws_ml.size(2);
   TORCH_CHECK(ws_acc.size(2) == num_chunks);
@@ -299,7 +299,7 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
                      (const ushort*)v_cache.data_ptr(),
                      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
                      (float)scale, G, Hkv, page_size, max_pages,
-                     (int)chunk_keys, num_chunks);
+                     (int)chunk_keys, num_chunks, (long long)q.stride(0));
   hipLaunchKernelGGL(paged_attention_reduce_kernel, dim3(B, Hq), dim3(128), 0,
                      stream, (ushort*)out.data_ptr(), ws_acc.data_ptr<float>(),
                      ws_ml.data_ptr<float>(), seq_lens.data_ptr<int>(),
@@ -311,7 +311,8 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v,
                        torch::Tensor slot_mapping) {
   TORCH_CHECK(k.is_cuda() && k.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == PA_HEAD_DIM);
+  TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == PA_HEAD_DIM);
   const int T = k.size(0);
   const int Hkv = k_cache.size(1);
   const int page_size = k_cache.size(2);
@@ -325,5 +326,6 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v,
                      (const ushort*)v.data_ptr(), (ushort*)k_cache.data_ptr(),
                      (ushort*)v_cache.data_ptr(),
                      (const long long*)slot_mapping.data_ptr<int64_t>(),
-                     T, Hkv, page_size);
+                     T, Hkv, page_size, (long long)k.stride(0),
+                     (long long)v.stride(0));
 }

@@ -1,15 +1,17 @@
 // Fused rotary position embedding (neox/Llama style) for q and k, gfx950.
 //
-// cos/sin are host-precomputed (fp32 table [max_pos, D/2] each, packed as
-// [max_pos, D] = cos|sin) per the CDNA4 guide's trig-table rule: on-device
-// sinf/cosf turns a memory-bound op VALU-bound (guide Appendix B).
+// cos/sin are host-precomputed (fp32 table [max_pos, D] = cos|sin halves)
+// per the CDNA4 guide's trig-table rule: on-device sinf/cosf turns a
+// memory-bound op VALU-bound (guide Appendix B).  q and k may be strided
+// row views straight into the fused qkv GEMM output (zero-copy decode).
 #include "common.h"
 
-// q: [T, Hq*D], k: [T, Hkv*D]; rotate pairs (i, i+D/2) within each head.
+// q: [T, Hq, D] rows at q_stride elems; k: [T, Hkv, D] rows at k_stride.
 __global__ void rope_kernel(ushort* __restrict__ q, ushort* __restrict__ k,
                             const float* __restrict__ cos_sin,
                             const int* __restrict__ positions,
-                            int T, int Hq, int Hkv, int D) {
+                            int T, int Hq, int Hkv, int D,
+                            long long q_stride, long long k_stride) {
   const int half = D / 2;
   const long long total = (long long)T * (Hq + Hkv) * half;
   long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -24,9 +26,9 @@ __global__ void rope_kernel(ushort* __restrict__ q, ushort* __restrict__ k,
     const float s = cos_sin[(long long)pos * D + half + i];
     ushort* base;
     if (h < Hq) {
-      base = q + ((long long)t * Hq + h) * D;
+      base = q + (long long)t * q_stride + (long long)h * D;
     } else {
-      base = k + ((long long)t * Hkv + (h - Hq)) * D;
+      base = k + (long long)t * k_stride + (long long)(h - Hq) * D;
     }
     float x1 = bf16_to_f32(base[i]);
     float x2 = bf16_to_f32(base[i + half]);
@@ -43,11 +45,20 @@ void rope(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32);
   TORCH_CHECK(positions.scalar_type() == torch::kInt32);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous());
   int T = q.size(0);
   int D = cos_sin.size(1);
-  TORCH_CHECK(q.numel() == (long long)T * num_q_heads * D, "rope: q shape");
-  TORCH_CHECK(k.numel() == (long long)T * num_kv_heads * D, "rope: k shape");
+  // accept [T, H*D] (stride(1)==1) or [T, H, D] views with contiguous rows
+  long long q_stride, k_stride;
+  if (q.dim() == 2) {
+    TORCH_CHECK(q.stride(1) == 1 && k.stride(1) == 1);
+    q_stride = q.stride(0);
+    k_stride = k.stride(0);
+  } else {
+    TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == D);
+    TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == D);
+    q_stride = q.stride(0);
+    k_stride = k.stride(0);
+  }
   long long total = (long long)T * (num_q_heads + num_kv_heads) * (D / 2);
   long long blocks = (total + 255) / 256;
   if (blocks > 4096) blocks = 4096;
@@ -56,5 +67,6 @@ void rope(torch::Tensor q, torch::Tensor k, torch::Tensor cos_sin,
   hipLaunchKernelGGL(rope_kernel, dim3((int)blocks), dim3(256), 0, stream,
                      (ushort*)q.data_ptr(), (ushort*)k.data_ptr(),
                      cos_sin.data_ptr<float>(), positions.data_ptr<int>(),
-                     T, (int)num_q_heads, (int)num_kv_heads, D);
+                     T, (int)num_q_heads, (int)num_kv_heads, D,
+                     q_stride, k_stride);
 }
