@@ -145,3 +145,25 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, slot_mapping: torch.Tensor) -> None:
     lib = require_native()
     lib.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+SKINNY_GEMM_MAX_M = 32
+
+
+def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
+                out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out[M,N] = x[M,K] @ w[N,K]^T for M <= 32 — split-K MFMA kernel that
+    fills all 256 CUs where hipBLASLt's heuristic tiles underfill the chip
+    on decode-shape GEMMs (see profiles/)."""
+    lib = require_native()
+    M, K = x.shape
+    N = w.size(0)
+    if out is None:
+        out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    n_blocks = (N + 63) // 64
+    split = min(max(1, 2048 // max(1, n_blocks)), max(1, K // 256))
+    k_slice = (K // split + 31) // 32 * 32
+    grid_y = (K + k_slice - 1) // k_slice
+    ws = torch.empty(grid_y, M, N, dtype=torch.float32, device=x.device)
+    lib.skinny_gemm(out, x, w, ws)
+    return out

@@ -17,6 +17,8 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 void reshape_and_cache(torch::Tensor k, torch::Tensor v,
                        torch::Tensor k_cache, torch::Tensor v_cache,
                        torch::Tensor slot_mapping);
+void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                 torch::Tensor ws);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, gfx950)");
@@ -28,4 +30,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "GQA paged attention decode with flash-decoding chunk split");
   m.def("reshape_and_cache", &reshape_and_cache,
         "scatter new k/v into the paged KV cache");
+  m.def("skinny_gemm", &skinny_gemm,
+        "split-K MFMA GEMM for decode-shape projections (M<=32)");
 }

@@ -105,6 +105,11 @@ class ShardedLinear:
         self.weight.normal_(0.0, std, generator=generator)
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        if x.is_cuda and x.dim() == 2 and x.size(0) <= 32:
+            # decode-shape path: hand-written split-K MFMA kernel (the
+            # hipBLASLt heuristic underfills the chip at M <= 32)
+            from .. import ops
+            return ops.skinny_gemm(x.contiguous(), self.weight)
         return x @ self.weight.t()
 
     @property

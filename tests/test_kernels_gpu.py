@@ -176,3 +176,18 @@ def test_decode_matches_prefill_math():
     out = ops.paged_attention_decode(q_full[:, :, -1].contiguous(), k_cache,
                                      v_cache, bt, seq_lens, scale)
     assert_close_bf16(out, ref, atol=3e-2, rtol=3e-2)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("M,N,K", [(1, 6144, 4096), (8, 4096, 4096),
+                                   (17, 1024, 2048), (32, 28672, 4096),
+                                   (32, 128256, 4096), (32, 4096, 14336)])
+def test_skinny_gemm(M, N, K):
+    import lws_amd.ops as ops
+
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.1
+    out = ops.skinny_gemm(x, w)
+    ref = (x.float() @ w.float().t()).to(torch.bfloat16)
+    assert_close_bf16(out, ref, atol=3e-2, rtol=3e-2)
