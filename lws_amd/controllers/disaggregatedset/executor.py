@@ -1,0 +1,228 @@
+"""Rolling-update executor: drives planner steps against the cluster.
+
+Behavioral port of reference pkg/controllers/disaggregatedset/executor.go:
+initRollingUpdate snapshots the initial-replicas annotation and creates
+0-replica new-revision LWS per role; ReconcileRollingUpdate gates on new-
+revision stability, computes the next planner step, scales up the new
+revision and drains old revisions newest-first with coordinated
+drain-to-zero; External roles have a no-shrink guard mid-rollout.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+from ...api import disaggregatedset as dsapi
+from ...api.meta import get_int_or_percent
+from ...utils import dsutils
+from .lws_manager import LeaderWorkerSetManager
+from .planner import (RollingUpdateConfig, compute_next_step,
+                      default_rolling_update_config)
+
+
+def _is_external(ds, role_name: str) -> bool:
+    for r in ds.spec.roles:
+        if r.name == role_name:
+            return dsapi.role_scaling_mode(r) == dsapi.RoleScalingMode.External
+    return False
+
+
+def get_target_replicas(ds, role_name: str, scalers: dict,
+                        current_new: int) -> int:
+    """executor.go:240-257."""
+    for r in ds.spec.roles:
+        if r.name != role_name:
+            continue
+        if dsapi.role_scaling_mode(r) == dsapi.RoleScalingMode.External:
+            s = scalers.get(role_name)
+            if s is not None:
+                return s.spec.replicas
+            return current_new
+        if r.spec.replicas is None:
+            return 1
+        return r.spec.replicas
+    return 1
+
+
+def extract_rolling_update_config(ds, all_role_names: list[str],
+                                  scalers: dict) -> list[RollingUpdateConfig]:
+    """executor.go:268-300 — percent scaling against the target replicas."""
+    config = default_rolling_update_config(len(all_role_names))
+    index = {n: i for i, n in enumerate(all_role_names)}
+    for role in ds.spec.roles:
+        rc = role.spec.rollout_strategy.rolling_update_configuration
+        if rc is None or role.name not in index:
+            continue
+        i = index[role.name]
+        replicas = get_target_replicas(ds, role.name, scalers, 0)
+        surge = get_int_or_percent(rc.max_surge if rc.max_surge is not None
+                                   else 0, replicas, True)
+        unavail = get_int_or_percent(
+            rc.max_unavailable if rc.max_unavailable is not None else 1,
+            replicas, False)
+        if unavail > 0:
+            config[i].max_unavailable = unavail
+            config[i].max_surge = surge
+        elif surge > 0:
+            config[i].max_surge = surge
+    return config
+
+
+def is_revision_stable(rev: dsutils.RevisionRoles,
+                       role_names: list[str]) -> bool:
+    """executor.go:310-321 — ReadyReplicas == Replicas on every role."""
+    for name in role_names:
+        lws = rev.roles.get(name)
+        if lws is None:
+            return False
+        if dsutils.get_lws_replicas(lws) != lws.status.ready_replicas:
+            return False
+    return True
+
+
+def _max_timestamp(rev: dsutils.RevisionRoles) -> float:
+    return max((lws.metadata.creation_timestamp or 0.0
+                for lws in rev.roles.values()), default=0.0)
+
+
+class RollingUpdateExecutor:
+    def __init__(self, lws_manager: LeaderWorkerSetManager, record=None):
+        self.lws_manager = lws_manager
+        self.record = record
+
+    # -- entry (executor.go:56-85) --------------------------------------
+    def reconcile_rolling_update_new(self, ds, slice_: int, revision: str,
+                                     scalers: dict) -> Optional[float]:
+        role_names = dsutils.get_role_names(ds)
+        role_configs = dsutils.get_role_configs(ds)
+        old_revisions, new_revision = \
+            self.lws_manager.get_revision_roles_list(ds, slice_, revision)
+        if not old_revisions:
+            return None
+        if new_revision is None:
+            return self._init_rolling_update(ds, slice_, revision, role_names,
+                                             role_configs, old_revisions)
+        return self._reconcile_rolling_update(ds, slice_, old_revisions,
+                                              new_revision, scalers)
+
+    def _init_rolling_update(self, ds, slice_, revision, role_names,
+                             role_configs, old_revisions) -> float:
+        """executor.go:87-127."""
+        for rev in old_revisions:
+            for lws in rev.roles.values():
+                replicas = dsutils.get_lws_replicas(lws)
+                self.lws_manager.set_initial_replicas(
+                    ds.metadata.namespace, lws.metadata.name, replicas)
+        for role_name in role_names:
+            self._ensure_new_lws_exists(ds, slice_, revision, role_name,
+                                        role_configs[role_name], 0)
+        return 1.0
+
+    def _ensure_new_lws_exists(self, ds, slice_, revision, role, config,
+                               initial_replicas: int) -> None:
+        lws_name = dsutils.generate_name(ds.metadata.name, slice_, revision,
+                                         role)
+        if self.lws_manager.get(ds, lws_name) is not None:
+            return
+        self.lws_manager.create(
+            ds, role=role, slice_=slice_, config=config, revision=revision,
+            labels=dsutils.generate_labels(ds.metadata.name, slice_, revision,
+                                           role),
+            replicas=initial_replicas)
+
+    # -- one step (executor.go:134-176) ----------------------------------
+    def _reconcile_rolling_update(self, ds, slice_, old_revisions,
+                                  new_revision, scalers) -> Optional[float]:
+        spec_role_names = dsutils.get_role_names(ds)
+        spec_set = set(spec_role_names)
+        old_set = {name for rev in old_revisions for name in rev.roles}
+        all_role_names = list(spec_role_names) + \
+            sorted(old_set - spec_set)
+
+        if not is_revision_stable(new_revision, spec_role_names):
+            return 1.0
+
+        initial_old, current_old, current_new, target_new = \
+            self._build_planner_state(ds, all_role_names, spec_set,
+                                      old_revisions, new_revision, scalers)
+        config = extract_rolling_update_config(ds, all_role_names, scalers)
+        step = compute_next_step(initial_old, current_old, current_new,
+                                 target_new, config)
+        if step is None:
+            return None  # rollout complete
+        self._scale_up_new(ds, slice_, new_revision, all_role_names, spec_set,
+                           current_new, step.new)
+        self._scale_down_old(ds, old_revisions, all_role_names, current_old,
+                             step.past)
+        return 0.2
+
+    def _build_planner_state(self, ds, all_role_names, spec_set,
+                             old_revisions, new_revision, scalers):
+        """executor.go:191-231."""
+        n = len(all_role_names)
+        initial_old = [0] * n
+        current_old = [0] * n
+        current_new = [0] * n
+        target_new = [0] * n
+        for i, name in enumerate(all_role_names):
+            initial_old[i] = old_revisions.total_initial_replicas(name)
+            current_old[i] = old_revisions.total_replicas(name)
+            if name in spec_set:
+                lws = new_revision.roles.get(name)
+                if lws is not None:
+                    current_new[i] = dsutils.get_lws_replicas(lws)
+                target_new[i] = get_target_replicas(ds, name, scalers,
+                                                    current_new[i])
+                # External no-shrink guard mid-rollout (executor.go:226-231)
+                if _is_external(ds, name) and len(old_revisions) > 0 and \
+                        target_new[i] < current_new[i]:
+                    target_new[i] = current_new[i]
+        return initial_old, current_old, current_new, target_new
+
+    def _scale_up_new(self, ds, slice_, new_revision, all_role_names,
+                      spec_set, current, target) -> None:
+        """executor.go:346-369."""
+        for i, name in enumerate(all_role_names):
+            if name not in spec_set or current[i] >= target[i]:
+                continue
+            lws_name = dsutils.generate_name(ds.metadata.name, slice_,
+                                             new_revision.revision, name)
+            self.lws_manager.scale(ds, lws_name, target[i])
+
+    def _scale_down_old(self, ds, old_revisions, role_names, current,
+                        target) -> None:
+        """executor.go:371-440 — newest-revision-first, with coordinated
+        drain-to-zero: when any role of a revision hits 0, the whole
+        revision drains to 0."""
+        budget = [current[i] - target[i] for i in range(len(role_names))]
+        for rev in sorted(old_revisions, key=_max_timestamp, reverse=True):
+            if all(b <= 0 for b in budget):
+                break
+            new_replicas: dict[str, int] = {}
+            planned_drain: dict[str, int] = {}
+            triggers: dict[str, bool] = {}
+            for i, name in enumerate(role_names):
+                lws = rev.roles.get(name)
+                if lws is None:
+                    continue
+                replicas = dsutils.get_lws_replicas(lws)
+                drain = min(max(0, budget[i]), replicas)
+                planned_drain[name] = drain
+                new_replicas[name] = replicas - drain
+                if new_replicas[name] == 0:
+                    triggers[name] = True
+            any_triggered = bool(triggers)
+            if any_triggered:
+                for name in role_names:
+                    if name in rev.roles:
+                        new_replicas[name] = 0
+            for i, name in enumerate(role_names):
+                lws = rev.roles.get(name)
+                if lws is None:
+                    continue
+                replicas = dsutils.get_lws_replicas(lws)
+                if replicas <= new_replicas[name]:
+                    continue
+                self.lws_manager.scale(ds, lws.metadata.name,
+                                       new_replicas[name])
+                if triggers.get(name) or not any_triggered:
+                    budget[i] -= planned_drain[name]
