@@ -148,6 +148,7 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor, k_cache: torch.Tensor,
 
 
 SKINNY_GEMM_MAX_M = 32
+_SKINNY_WS: dict = {}
 
 
 def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
@@ -160,10 +161,16 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
     N = w.size(0)
     if out is None:
         out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    import os
+    target = int(os.environ.get("LWS_SG_TARGET", "1024"))
     n_blocks = (N + 63) // 64
-    split = min(max(1, 2048 // max(1, n_blocks)), max(1, K // 256))
-    k_slice = (K // split + 31) // 32 * 32
+    split = min(max(1, target // max(1, n_blocks)), max(1, K // 128))
+    k_slice = (K // split + 127) // 128 * 128
     grid_y = (K + k_slice - 1) // k_slice
-    ws = torch.empty(grid_y, M, N, dtype=torch.float32, device=x.device)
+    key = (grid_y, M, N, x.device.index)
+    ws = _SKINNY_WS.get(key)
+    if ws is None:
+        ws = torch.empty(grid_y, M, N, dtype=torch.float32, device=x.device)
+        _SKINNY_WS[key] = ws
     lib.skinny_gemm(out, x, w, ws)
     return out

@@ -1,22 +1,31 @@
 #include "hip/hip_runtime.h"
-// Split-K MFMA "skinny" GEMM for decode-shape projections on gfx950.
+// Split-K MFMA "skinny" GEMM for decode-shape projections on gfx950. (v3)
 //
 // Motivation (profiles/r01_decode8b_kernel_stats.md): at decode batch
-// M<=64, hipBLASLt's heuristic tiles launch ~48 workgroups for e.g.
-// [32,4096]x[4096,6144] — 48 of 256 CUs — and read weights at ~1 TB/s
-// cold.  This kernel splits K so the launch fills the chip and streams the
-// weight matrix at HBM rate; the tiny activation matrix stays L2-resident.
+// M<=32, hipBLASLt's heuristic tiles underfill the 256-CU chip and read
+// cold weights at ~1 TB/s.  This kernel splits K so the launch fills the
+// chip and streams W at HBM rate.
 //
 //   out[M, N] = x[M, K] @ W[N, K]^T      (bf16 in, fp32 accumulate)
 //
-// Geometry: one workgroup = 4 waves, each wave owns one 16-wide n-tile
-// (64 n per workgroup) and a K-slice; per K-step each wave issues
-// ceil(M/16) mfma_f32_16x16x32_bf16.  Each K-slice writes a partial fp32
-// plane; a finalize kernel reduces the planes and converts to bf16
-// (deterministic split-K, no atomics).
+// v3 design (CDNA4 guide §5 + technique catalog):
+//  - async global->LDS staging via __builtin_amdgcn_global_load_lds
+//    (16 B/lane direct DMA, no VGPR round trip — Common-mistake #1)
+//  - double-buffered K sub-slices: next tile's loads issue BEFORE the
+//    current tile's MFMA loop, the trailing __syncthreads drains them
+//    (minimum 2-phase pattern, guide §5.5 T3 recipe)
+//  - LDS destination of global_load_lds is linear (wave-uniform base +
+//    lane*16), so bank-conflict avoidance uses the XOR swizzle applied to
+//    BOTH the per-lane global source address and the ds_read offset —
+//    same involution on both sides (guide ERRATA #21 / T2)
+//  - one workgroup = 4 waves; each wave owns a 16-wide n-tile (64 n per
+//    block) and a K-slice; per 32-k step each wave issues ceil(M/16)
+//    mfma_f32_16x16x32_bf16
+//  - deterministic split-K: per-slice fp32 partial planes + a finalize
+//    reduce kernel (no atomics)
 //
 // Fragment layouts (verified on MI355X by tests/test_kernels_gpu.py
-// numerics + the asymmetric-input rule from the CDNA4 guide §3):
+// numerics with asymmetric inputs — guide §3 rule):
 //   A frag: lane l holds A[m = l%16][k = (l/16)*8 + j], j=0..7
 //   B frag: lane l holds B[k = (l/16)*8 + j][n = l%16]
 //   C/D:    lane l reg r holds D[row = (l/16)*4 + r][col = l%16]
@@ -27,12 +36,39 @@ using f32x4_t = __attribute__((ext_vector_type(4))) float;
 
 #define SG_NTILE 16
 #define SG_WAVES 4
+#define SG_ROWS (SG_NTILE * SG_WAVES)   // 64 W rows per block
+#define SG_KSUB 128                     // k elems per sub-slice (256 B/row)
+#define SG_ROWB (SG_KSUB * 2)           // bytes per LDS row
 
-// Staged K sub-slice width (elements) and padded LDS row pitch.  The +8
-// element pad (16 B) breaks the 512 B power-of-2 row stride that would put
-// all 16 fragment lanes in the same LDS bank (guide §6 Guideline 4).
-#define SG_KSUB 256
-#define SG_LDS_PITCH (SG_KSUB + 8)
+typedef __attribute__((address_space(3))) uint32_t lds_u32;
+typedef __attribute__((address_space(1))) const uint32_t glb_u32;
+
+// XOR swizzle inside a row: flips byte-addr bits 4..6 by the low row bits,
+// spreading the 16 fragment lanes (which read the same 16 B column range
+// of 16 different rows) across 8 distinct banks (2-way is free — G4).
+__device__ __forceinline__ int sg_swz(int row, int colb) {
+  return colb ^ ((row & 7) << 4);
+}
+
+// Stage a [rows x SG_KSUB] bf16 tile into linear LDS via global_load_lds.
+// Per wave instruction: 64 lanes x 16 B = 1 KiB = 4 LDS rows.  The global
+// source address carries the inverse swizzle so a swizzled ds_read
+// recovers the logical element (write-side linear, source+read swizzled).
+__device__ __forceinline__ void sg_stage_async(
+    ushort* lds_tile, const ushort* src_base, long long src_row_stride,
+    int rows, int src_row_limit, int wave, int lane) {
+  const int nunits = rows * SG_ROWB / 1024;
+  for (int u = wave; u < nunits; u += SG_WAVES) {
+    const int lb = u * 1024 + lane * 16;
+    int row = lb >> 8;
+    const int colb = sg_swz(row, lb & 255);
+    if (row >= src_row_limit) row = src_row_limit - 1;  // clamped, unused
+    const ushort* src = src_base + (long long)row * src_row_stride
+        + (colb >> 1);
+    lds_u32* dst = (lds_u32*)(lds_tile + u * 512);  // wave-uniform base
+    __builtin_amdgcn_global_load_lds((glb_u32*)src, dst, 16, 0, 0);
+  }
+}
 
 template <int MTILES>
 __global__ __launch_bounds__(256)
@@ -42,46 +78,52 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
                         int M, int N, int K, int k_slice) {
   const int wave = threadIdx.x / WAVE_SIZE;
   const int lane = threadIdx.x % WAVE_SIZE;
-  const int n0 = blockIdx.x * (SG_NTILE * SG_WAVES);
+  const int n0 = blockIdx.x * SG_ROWS;
   const int kbegin = blockIdx.y * k_slice;
   const int kend = min(kbegin + k_slice, K);
 
   const int frag_row = lane % 16;        // m (A) / n (B)
   const int frag_kgrp = lane / 16;       // which 8-wide k group
 
-  // W tile [64 n x SG_KSUB k] staged cooperatively (coalesced 64 B row
-  // chunks from HBM), fragments then read via ds_read_b128.
-  __shared__ ushort w_lds[SG_NTILE * SG_WAVES][SG_LDS_PITCH];
+  __shared__ ushort w_lds[2][SG_ROWS * SG_KSUB];
+  __shared__ ushort x_lds[2][16 * MTILES * SG_KSUB];
 
   f32x4_t acc[MTILES];
 #pragma unroll
   for (int t = 0; t < MTILES; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
 
-  for (int ks0 = kbegin; ks0 < kend; ks0 += SG_KSUB) {
-    const int kw = min(SG_KSUB, kend - ks0);      // valid k width (mult of 32)
-    // ---- stage: 64 rows x kw elems; consecutive threads take consecutive
-    // 8-elem units within a row -> fully coalesced global reads
-    const int units_per_row = kw / 8;
-    const int total_units = (SG_NTILE * SG_WAVES) * units_per_row;
-    __syncthreads();
-    for (int u = threadIdx.x; u < total_units; u += blockDim.x) {
-      const int row = u / units_per_row;
-      const int kc = (u % units_per_row) * 8;
-      uint4 val = make_uint4(0, 0, 0, 0);
-      if (n0 + row < N)
-        val = *reinterpret_cast<const uint4*>(
-            w + (long long)(n0 + row) * K + ks0 + kc);
-      *reinterpret_cast<uint4*>(&w_lds[row][kc]) = val;
-    }
-    __syncthreads();
+  const ushort* w_base = w + (long long)n0 * K;
+  const int w_rows_valid = min(SG_ROWS, N - n0);
+  const int nsub = (kend - kbegin) / SG_KSUB;
 
-    for (int k0 = 0; k0 < kw; k0 += 32) {
-      const int kf = ks0 + k0 + frag_kgrp * 8;
+  // prologue: stage sub-slice 0 into buffer 0
+  sg_stage_async(w_lds[0], w_base + kbegin, K, SG_ROWS, w_rows_valid, wave,
+                 lane);
+  sg_stage_async(x_lds[0], x + kbegin, K, 16 * MTILES, M, wave, lane);
+  __syncthreads();
+
+  for (int s = 0; s < nsub; ++s) {
+    const int cur = s & 1;
+    if (s + 1 < nsub) {
+      const int ks_next = kbegin + (s + 1) * SG_KSUB;
+      sg_stage_async(w_lds[cur ^ 1], w_base + ks_next, K, SG_ROWS,
+                     w_rows_valid, wave, lane);
+      sg_stage_async(x_lds[cur ^ 1], x + ks_next, K, 16 * MTILES, M, wave,
+                     lane);
+    }
+    // compute current sub-slice: 4 k-steps of 32
+    const ushort* wt = w_lds[cur];
+    const ushort* xt = x_lds[cur];
+#pragma unroll
+    for (int k0 = 0; k0 < SG_KSUB; k0 += 32) {
+      const int colb = (k0 + frag_kgrp * 8) * 2;
       bf16x8_t bfrag;
       {
+        const int row = wave * SG_NTILE + frag_row;
         bf16x8 tmp;
         tmp.u = *reinterpret_cast<const uint4*>(
-            &w_lds[wave * SG_NTILE + frag_row][k0 + frag_kgrp * 8]);
+            reinterpret_cast<const char*>(wt) + row * SG_ROWB +
+            sg_swz(row, colb));
 #pragma unroll
         for (int j = 0; j < 8; ++j) bfrag[j] = (short)tmp.h[j];
       }
@@ -89,25 +131,24 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
       for (int t = 0; t < MTILES; ++t) {
         const int m = t * 16 + frag_row;
         bf16x8_t afrag;
-        if (m < M) {
-          // x is tiny (M<=32 rows) and L2-resident; strided reads are cheap
+        {
           bf16x8 tmp;
-          tmp.u = *reinterpret_cast<const uint4*>(x + (long long)m * K + kf);
+          tmp.u = *reinterpret_cast<const uint4*>(
+              reinterpret_cast<const char*>(xt) + m * SG_ROWB +
+              sg_swz(m, colb));
 #pragma unroll
           for (int j = 0; j < 8; ++j) afrag[j] = (short)tmp.h[j];
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) afrag[j] = 0;
         }
         acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[t],
                                                          0, 0, 0);
       }
     }
+    // drains the in-flight global_load_lds of the next buffer AND fences
+    // the current buffer for reuse (minimum 2-phase schedule)
+    __syncthreads();
   }
 
   // C/D layout: lane l reg r -> D[row=(l/16)*4+r][col=l%16].
-  // Deterministic split-K: each k-slice writes its own partial plane;
-  // the finalize kernel reduces over slices (no atomics).
   const int n = n0 + wave * SG_NTILE + frag_row;
   float* plane = out_ws + (long long)blockIdx.y * M * N;
   if (n < N) {
@@ -138,6 +179,11 @@ __global__ void skinny_gemm_finalize_kernel(ushort* __restrict__ out,
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+static int env_int(const char* name, int dflt) {
+  const char* v = getenv(name);
+  return v ? atoi(v) : dflt;
+}
+
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
@@ -148,16 +194,19 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const int N = w.size(0);
   TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
   TORCH_CHECK(M <= 32, "skinny_gemm: M must be <= 32");
-  TORCH_CHECK(K % 32 == 0 && N % 16 == 0);
+  TORCH_CHECK(K % SG_KSUB == 0 && N % 16 == 0,
+              "skinny_gemm: K must be a multiple of 128");
 
   hipStream_t stream = at::hip::getCurrentHIPStream();
-  const int n_blocks = (N + SG_NTILE * SG_WAVES - 1) / (SG_NTILE * SG_WAVES);
-  // split K so the grid lands near ~2048 workgroups (8 per CU)
-  int split = 2048 / max(1, n_blocks);
-  const int max_split = max(1, K / 256);
+  const int n_blocks = (N + SG_ROWS - 1) / SG_ROWS;
+  // split K so the grid lands near the target workgroup count —
+  // balances chip fill against split-K workspace traffic
+  const int target = env_int("LWS_SG_TARGET", 1024);
+  int split = target / max(1, n_blocks);
+  const int max_split = max(1, K / SG_KSUB);
   if (split > max_split) split = max_split;
   if (split < 1) split = 1;
-  int k_slice = (K / split + 31) / 32 * 32;
+  int k_slice = (K / split + SG_KSUB - 1) / SG_KSUB * SG_KSUB;
   const int grid_y = (K + k_slice - 1) / k_slice;
   TORCH_CHECK(ws.numel() >= (long long)grid_y * M * N,
               "skinny_gemm workspace too small");

@@ -105,9 +105,13 @@ class ShardedLinear:
         self.weight.normal_(0.0, std, generator=generator)
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
-        if x.is_cuda and x.dim() == 2 and x.size(0) <= 32:
-            # decode-shape path: hand-written split-K MFMA kernel (the
-            # hipBLASLt heuristic underfills the chip at M <= 32)
+        if x.is_cuda and x.dim() == 2 and x.size(0) <= 32 and \
+                x.size(1) % 128 == 0 and (self.weight.numel() <= 32 * 1024 * 1024
+                                          or self.weight.size(0) >= 65536):
+            # decode-shape path for small weight shards (<=64 MB bf16, the
+            # TP-sharded regime): hand-written split-K MFMA kernel — the
+            # hipBLASLt heuristic underfills the chip there (see profiles/).
+            # Large shards stay on hipBLASLt, which fills the chip fine.
             from .. import ops
             return ops.skinny_gemm(x.contiguous(), self.weight)
         return x @ self.weight.t()
