@@ -1,0 +1,96 @@
+"""lws_amd controller-manager entrypoint (cmd/main.go equivalent).
+
+    python -m lws_amd --config config.yaml
+
+Boots the full standalone control plane: object store + admission,
+StatefulSet controller, scheduler over the configured node inventory,
+node agents, LWS/Pod/DS reconcilers, gang provider, and the HTTP API
+server with healthz/readyz/metrics.  Deprecated-style CLI flags override
+the config file when explicitly set (cmd/main.go:124-126 behavior).
+"""
+from __future__ import annotations
+
+import argparse
+import logging
+import signal
+import sys
+import time
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(prog="lws-amd-manager")
+    p.add_argument("--config", help="Configuration YAML path")
+    p.add_argument("--api-bind", help="override apiServer.bindAddress")
+    p.add_argument("--scheduler-provider",
+                   help="override gangScheduling.schedulerProvider")
+    p.add_argument("--nodes", type=int,
+                   help="override: N uniform nodes with 8 GPUs each")
+    p.add_argument("--gpus-per-node", type=int, default=8)
+    p.add_argument("--zap-log-level", default="info")
+    args = p.parse_args(argv)
+
+    logging.basicConfig(
+        level=getattr(logging, args.zap_log_level.upper(), logging.INFO),
+        format="%(asctime)s %(levelname)s %(name)s %(message)s")
+    log = logging.getLogger("lws_amd.manager")
+
+    from . import config as cfgmod
+    cfg = cfgmod.load(args.config) if args.config else cfgmod.Configuration()
+    if args.api_bind:
+        cfg.api_server.bind_address = args.api_bind
+    if args.scheduler_provider is not None:
+        cfg.gang_scheduling.scheduler_provider = args.scheduler_provider
+        cfgmod.validate(cfg)
+
+    from .api.meta import ObjectMeta
+    from .cluster.cluster import LwsCluster, make_nodes
+    from .cluster.node import Node
+
+    if args.nodes:
+        nodes = make_nodes(args.nodes, gpus_per_node=args.gpus_per_node,
+                           topology_key=cfg.topology_key)
+    elif cfg.nodes:
+        nodes = []
+        for nc in cfg.nodes:
+            n = Node(address=nc.address)
+            n.metadata = ObjectMeta(name=nc.name)
+            n.metadata.labels = {"kubernetes.io/hostname": nc.name,
+                                 **nc.labels}
+            n.capacity = {"amd.com/gpu": nc.gpus}
+            nodes.append(n)
+    else:
+        nodes = make_nodes(1, topology_key=cfg.topology_key)
+
+    provider_factory = None
+    if cfg.gang_scheduling.scheduler_provider:
+        from .schedulerprovider.provider import new_scheduler_provider
+        name = cfg.gang_scheduling.scheduler_provider
+        provider_factory = lambda store: new_scheduler_provider(name, store)  # noqa: E731
+
+    cluster = LwsCluster(nodes=nodes,
+                         scheduler_provider_factory=provider_factory).start()
+    log.info("controller manager started with %d nodes", len(nodes))
+
+    server = None
+    if cfg.api_server.enable:
+        from .apiserver import ApiServer
+        server = ApiServer(cluster.store, cfg.api_server.bind_address)
+        server.start()
+        log.info("api server listening on %s", cfg.api_server.bind_address)
+
+    stop = []
+    signal.signal(signal.SIGTERM, lambda *a: stop.append(1))
+    signal.signal(signal.SIGINT, lambda *a: stop.append(1))
+    try:
+        while not stop:
+            time.sleep(0.2)
+    finally:
+        if server is not None:
+            server.stop()
+        cluster.stop()
+        log.info("controller manager stopped")
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,195 @@
+"""HTTP API server exposing the lws_amd object store.
+
+The reference delegates its API surface to kube-apiserver; standalone
+lws_amd serves the same verbs itself (fastapi): apply/get/list/delete of
+LeaderWorkerSet, DisaggregatedSet, DisaggregatedSetRoleScaler, plus
+read-only Pods/StatefulSets/Services/PodGroups, the /scale subresource on
+scalers (HPA-compatible shape), healthz/readyz and Prometheus-style
+/metrics.  Admission (defaulting + validation webhooks) runs in-process on
+every write, exactly as registered on the store.
+"""
+from __future__ import annotations
+
+import threading
+import time
+from typing import Any, Optional
+
+from .api import serde
+from .api import disaggregatedset as dsapi
+from .api import leaderworkerset as lwsapi
+from .api.disaggregatedset import DisaggregatedSet, DisaggregatedSetRoleScaler
+from .api.leaderworkerset import LeaderWorkerSet
+from .cluster.store import ApiError, NotFoundError, Store
+
+KIND_MODELS = {
+    "leaderworkersets": (lwsapi.KIND, LeaderWorkerSet),
+    "disaggregatedsets": (dsapi.KIND, DisaggregatedSet),
+    "disaggregatedsetrolescalers": (dsapi.SCALER_KIND,
+                                    DisaggregatedSetRoleScaler),
+}
+READONLY_KINDS = {
+    "pods": "Pod",
+    "statefulsets": "StatefulSet",
+    "services": "Service",
+    "podgroups": "PodGroup",
+    "controllerrevisions": "ControllerRevision",
+}
+
+
+def build_app(store: Store, manager=None):
+    from fastapi import FastAPI, HTTPException
+
+    app = FastAPI(title="lws-amd", version="0.1")
+    started = time.time()
+
+    def _err(e: ApiError):
+        code = {"NotFound": 404, "Conflict": 409, "AlreadyExists": 409,
+                "Invalid": 422}.get(e.reason, 400)
+        raise HTTPException(status_code=code, detail=str(e))
+
+    @app.get("/healthz")
+    def healthz():
+        return {"status": "ok"}
+
+    @app.get("/readyz")
+    def readyz():
+        return {"status": "ok", "uptime_s": round(time.time() - started, 1)}
+
+    @app.get("/metrics")
+    def metrics():
+        from fastapi.responses import PlainTextResponse
+
+        lines = ["# TYPE lws_amd_objects gauge"]
+        counts: dict[str, int] = {}
+        for kind, ns, name in store.snapshot_keys():
+            counts[kind] = counts.get(kind, 0) + 1
+        for kind, n in sorted(counts.items()):
+            lines.append(f'lws_amd_objects{{kind="{kind}"}} {n}')
+        return PlainTextResponse("\n".join(lines) + "\n")
+
+    @app.post("/apis/{resource}/namespaces/{ns}")
+    def create(resource: str, ns: str, body: dict):
+        if resource not in KIND_MODELS:
+            raise HTTPException(404, f"unknown resource {resource}")
+        kind, model = KIND_MODELS[resource]
+        obj = serde.from_dict(model, body)
+        obj.metadata.namespace = ns
+        try:
+            return serde.to_dict(store.create(obj))
+        except ApiError as e:
+            _err(e)
+
+    @app.put("/apis/{resource}/namespaces/{ns}/{name}")
+    def apply(resource: str, ns: str, name: str, body: dict):
+        if resource not in KIND_MODELS:
+            raise HTTPException(404, f"unknown resource {resource}")
+        kind, model = KIND_MODELS[resource]
+        obj = serde.from_dict(model, body)
+        obj.metadata.namespace = ns
+        obj.metadata.name = name
+        existing = store.try_get(kind, ns, name)
+        try:
+            if existing is None:
+                return serde.to_dict(store.create(obj))
+            existing.spec = obj.spec
+            existing.metadata.labels = obj.metadata.labels
+            existing.metadata.annotations = obj.metadata.annotations
+            return serde.to_dict(store.update(existing))
+        except ApiError as e:
+            _err(e)
+
+    @app.get("/apis/{resource}/namespaces/{ns}")
+    def list_(resource: str, ns: str):
+        kind = (KIND_MODELS.get(resource) or (None,))[0] or \
+            READONLY_KINDS.get(resource)
+        if kind is None:
+            raise HTTPException(404, f"unknown resource {resource}")
+        return {"items": [serde.to_dict(o) for o in store.list(kind, ns)]}
+
+    @app.get("/apis/{resource}/namespaces/{ns}/{name}")
+    def get(resource: str, ns: str, name: str):
+        kind = (KIND_MODELS.get(resource) or (None,))[0] or \
+            READONLY_KINDS.get(resource)
+        if kind is None:
+            raise HTTPException(404, f"unknown resource {resource}")
+        obj = store.try_get(kind, ns, name)
+        if obj is None:
+            raise HTTPException(404, f"{kind} {ns}/{name} not found")
+        return serde.to_dict(obj)
+
+    @app.delete("/apis/{resource}/namespaces/{ns}/{name}")
+    def delete(resource: str, ns: str, name: str,
+               propagation: str = "Background"):
+        kind = (KIND_MODELS.get(resource) or (None,))[0]
+        if kind is None:
+            raise HTTPException(404, f"unknown resource {resource}")
+        try:
+            store.delete(kind, ns, name, propagation=propagation)
+        except ApiError as e:
+            _err(e)
+        return {"status": "deleted"}
+
+    # /scale subresource (HPA-compatible shape) on LWS + RoleScaler
+    @app.get("/apis/{resource}/namespaces/{ns}/{name}/scale")
+    def get_scale(resource: str, ns: str, name: str):
+        if resource == "leaderworkersets":
+            obj = store.try_get(lwsapi.KIND, ns, name)
+            if obj is None:
+                raise HTTPException(404, "not found")
+            return {"spec": {"replicas": obj.spec.replicas},
+                    "status": {"replicas": obj.status.replicas,
+                               "selector": obj.status.hpa_pod_selector}}
+        if resource == "disaggregatedsetrolescalers":
+            obj = store.try_get(dsapi.SCALER_KIND, ns, name)
+            if obj is None:
+                raise HTTPException(404, "not found")
+            return {"spec": {"replicas": obj.spec.replicas},
+                    "status": {"replicas": obj.status.replicas,
+                               "selector": obj.status.selector}}
+        raise HTTPException(404, f"no scale subresource on {resource}")
+
+    @app.put("/apis/{resource}/namespaces/{ns}/{name}/scale")
+    def put_scale(resource: str, ns: str, name: str, body: dict):
+        replicas = int(body.get("spec", {}).get("replicas", 0))
+        kind = {"leaderworkersets": lwsapi.KIND,
+                "disaggregatedsetrolescalers": dsapi.SCALER_KIND}.get(resource)
+        if kind is None:
+            raise HTTPException(404, f"no scale subresource on {resource}")
+        obj = store.try_get(kind, ns, name)
+        if obj is None:
+            raise HTTPException(404, "not found")
+        obj.spec.replicas = replicas
+        try:
+            store.update(obj)
+        except ApiError as e:
+            _err(e)
+        return get_scale(resource, ns, name)
+
+    return app
+
+
+class ApiServer:
+    """Uvicorn server on a background thread."""
+
+    def __init__(self, store: Store, bind: str = "127.0.0.1:8080"):
+        host, _, port = bind.rpartition(":")
+        self.host = host.lstrip(":") or "127.0.0.1"
+        self.port = int(port)
+        self.app = build_app(store)
+        self._server = None
+        self._thread: Optional[threading.Thread] = None
+
+    def start(self) -> None:
+        import uvicorn
+
+        config = uvicorn.Config(self.app, host=self.host, port=self.port,
+                                log_level="warning")
+        self._server = uvicorn.Server(config)
+        self._thread = threading.Thread(target=self._server.run, daemon=True)
+        self._thread.start()
+
+    def stop(self) -> None:
+        if self._server is not None:
+            self._server.should_exit = True
+        if self._thread is not None:
+            self._thread.join(timeout=5)

@@ -1,0 +1,276 @@
+"""Typed client for the lws_amd API (client-go clientset equivalent).
+
+Two transports behind one typed surface (reference client-go generates
+clientset/informers/listers — SURVEY.md §2.8):
+ - HTTP: talks to the lws_amd API server (python -m lws_amd)
+ - in-process: wraps a Store directly (fake-clientset analogue for tests)
+
+Typed accessors: leader_worker_sets / disaggregated_sets / role_scalers
+(+ read-only pods/statefulsets/services) with create/get/list/update/
+delete/scale, and Informer/Lister helpers for cached watching.
+"""
+from __future__ import annotations
+
+import threading
+import time
+from typing import Any, Callable, Optional
+
+from ..api import serde
+from ..api import disaggregatedset as dsapi
+from ..api import leaderworkerset as lwsapi
+from ..api.core import Pod, Service, StatefulSet
+from ..api.disaggregatedset import DisaggregatedSet, DisaggregatedSetRoleScaler
+from ..api.leaderworkerset import LeaderWorkerSet
+
+RESOURCES = {
+    "leaderworkersets": (lwsapi.KIND, LeaderWorkerSet),
+    "disaggregatedsets": (dsapi.KIND, DisaggregatedSet),
+    "disaggregatedsetrolescalers": (dsapi.SCALER_KIND,
+                                    DisaggregatedSetRoleScaler),
+    "pods": ("Pod", Pod),
+    "statefulsets": ("StatefulSet", StatefulSet),
+    "services": ("Service", Service),
+}
+KIND_TO_RESOURCE = {kind: res for res, (kind, _) in RESOURCES.items()}
+
+
+class ResourceClient:
+    def __init__(self, transport, resource: str, namespace: str = "default"):
+        self.transport = transport
+        self.resource = resource
+        self.namespace = namespace
+        self.kind, self.model = RESOURCES[resource]
+
+    def create(self, obj):
+        return self.transport.create(self, obj)
+
+    def apply(self, obj):
+        return self.transport.apply(self, obj)
+
+    def get(self, name: str):
+        return self.transport.get(self, name)
+
+    def list(self):
+        return self.transport.list(self)
+
+    def update(self, obj):
+        return self.transport.update(self, obj)
+
+    def delete(self, name: str, propagation: str = "Background"):
+        return self.transport.delete(self, name, propagation)
+
+    def scale(self, name: str, replicas: int):
+        return self.transport.scale(self, name, replicas)
+
+    def get_scale(self, name: str):
+        return self.transport.get_scale(self, name)
+
+
+class StoreTransport:
+    """In-process transport over a Store (fake-clientset analogue)."""
+
+    def __init__(self, store):
+        self.store = store
+
+    def create(self, rc, obj):
+        obj.metadata.namespace = obj.metadata.namespace or rc.namespace
+        return self.store.create(obj)
+
+    def apply(self, rc, obj):
+        obj.metadata.namespace = obj.metadata.namespace or rc.namespace
+        existing = self.store.try_get(rc.kind, obj.metadata.namespace,
+                                      obj.metadata.name)
+        if existing is None:
+            return self.store.create(obj)
+        existing.spec = obj.spec
+        existing.metadata.labels = obj.metadata.labels
+        existing.metadata.annotations = obj.metadata.annotations
+        return self.store.update(existing)
+
+    def get(self, rc, name):
+        return self.store.try_get(rc.kind, rc.namespace, name)
+
+    def list(self, rc):
+        return self.store.list(rc.kind, rc.namespace)
+
+    def update(self, rc, obj):
+        return self.store.update(obj)
+
+    def delete(self, rc, name, propagation):
+        from ..cluster.store import NotFoundError
+        try:
+            self.store.delete(rc.kind, rc.namespace, name,
+                              propagation=propagation)
+        except NotFoundError:
+            pass
+
+    def scale(self, rc, name, replicas):
+        obj = self.store.get(rc.kind, rc.namespace, name)
+        obj.spec.replicas = replicas
+        return self.store.update(obj)
+
+    def get_scale(self, rc, name):
+        obj = self.store.get(rc.kind, rc.namespace, name)
+        sel = getattr(obj.status, "hpa_pod_selector",
+                      getattr(obj.status, "selector", ""))
+        return {"spec": {"replicas": obj.spec.replicas},
+                "status": {"replicas": obj.status.replicas, "selector": sel}}
+
+
+class HttpTransport:
+    """HTTP transport against the lws_amd API server."""
+
+    def __init__(self, base_url: str):
+        import httpx
+
+        self.base = base_url.rstrip("/")
+        self.http = httpx.Client(timeout=30)
+
+    def _url(self, rc, name: Optional[str] = None, sub: str = ""):
+        u = f"{self.base}/apis/{rc.resource}/namespaces/{rc.namespace}"
+        if name:
+            u += f"/{name}"
+        return u + sub
+
+    def _obj(self, rc, data):
+        return serde.from_dict(rc.model, data)
+
+    def create(self, rc, obj):
+        r = self.http.post(self._url(rc), json=serde.to_dict(obj))
+        r.raise_for_status()
+        return self._obj(rc, r.json())
+
+    def apply(self, rc, obj):
+        r = self.http.put(self._url(rc, obj.metadata.name),
+                          json=serde.to_dict(obj))
+        r.raise_for_status()
+        return self._obj(rc, r.json())
+
+    def get(self, rc, name):
+        r = self.http.get(self._url(rc, name))
+        if r.status_code == 404:
+            return None
+        r.raise_for_status()
+        return self._obj(rc, r.json())
+
+    def list(self, rc):
+        r = self.http.get(self._url(rc))
+        r.raise_for_status()
+        return [self._obj(rc, item) for item in r.json()["items"]]
+
+    def update(self, rc, obj):
+        return self.apply(rc, obj)
+
+    def delete(self, rc, name, propagation):
+        r = self.http.delete(self._url(rc, name),
+                             params={"propagation": propagation})
+        if r.status_code not in (200, 404):
+            r.raise_for_status()
+
+    def scale(self, rc, name, replicas):
+        r = self.http.put(self._url(rc, name, "/scale"),
+                          json={"spec": {"replicas": replicas}})
+        r.raise_for_status()
+        return r.json()
+
+    def get_scale(self, rc, name):
+        r = self.http.get(self._url(rc, name, "/scale"))
+        r.raise_for_status()
+        return r.json()
+
+    def healthz(self) -> bool:
+        try:
+            return self.http.get(f"{self.base}/healthz").status_code == 200
+        except Exception:  # noqa: BLE001
+            return False
+
+
+class Clientset:
+    """Typed accessor bundle (client-go Clientset equivalent)."""
+
+    def __init__(self, transport):
+        self.transport = transport
+
+    @classmethod
+    def for_server(cls, base_url: str) -> "Clientset":
+        return cls(HttpTransport(base_url))
+
+    @classmethod
+    def for_store(cls, store) -> "Clientset":
+        return cls(StoreTransport(store))
+
+    def leader_worker_sets(self, namespace="default") -> ResourceClient:
+        return ResourceClient(self.transport, "leaderworkersets", namespace)
+
+    def disaggregated_sets(self, namespace="default") -> ResourceClient:
+        return ResourceClient(self.transport, "disaggregatedsets", namespace)
+
+    def role_scalers(self, namespace="default") -> ResourceClient:
+        return ResourceClient(self.transport, "disaggregatedsetrolescalers",
+                              namespace)
+
+    def pods(self, namespace="default") -> ResourceClient:
+        return ResourceClient(self.transport, "pods", namespace)
+
+    def statefulsets(self, namespace="default") -> ResourceClient:
+        return ResourceClient(self.transport, "statefulsets", namespace)
+
+    def services(self, namespace="default") -> ResourceClient:
+        return ResourceClient(self.transport, "services", namespace)
+
+
+class Informer:
+    """Polling informer with a local cache + event callbacks (informer/
+    lister equivalent for the HTTP transport; in-process users can watch
+    the store directly)."""
+
+    def __init__(self, rc: ResourceClient, resync_seconds: float = 1.0):
+        self.rc = rc
+        self.resync = resync_seconds
+        self.cache: dict[str, Any] = {}
+        self.handlers: list[Callable[[str, Any], None]] = []
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+
+    def add_handler(self, fn: Callable[[str, Any], None]) -> None:
+        self.handlers.append(fn)
+
+    def lister(self) -> list:
+        return list(self.cache.values())
+
+    def start(self) -> "Informer":
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+        return self
+
+    def _run(self) -> None:
+        while not self._stop.is_set():
+            try:
+                items = {o.metadata.name: o for o in self.rc.list()}
+            except Exception:  # noqa: BLE001
+                time.sleep(self.resync)
+                continue
+            for name, obj in items.items():
+                old = self.cache.get(name)
+                if old is None:
+                    self._emit("ADDED", obj)
+                elif old.metadata.resource_version != \
+                        obj.metadata.resource_version:
+                    self._emit("MODIFIED", obj)
+            for name in list(self.cache):
+                if name not in items:
+                    self._emit("DELETED", self.cache[name])
+            self.cache = items
+            self._stop.wait(self.resync)
+
+    def _emit(self, event: str, obj) -> None:
+        for fn in self.handlers:
+            try:
+                fn(event, obj)
+            except Exception:  # noqa: BLE001
+                pass
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread:
+            self._thread.join(timeout=5)

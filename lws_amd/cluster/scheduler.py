@@ -71,7 +71,9 @@ class Scheduler:
 
         for gname, pods in sorted(gangs.items()):
             pg = self.store.try_get("PodGroup", namespace, gname)
-            min_member = pg.spec.min_member if pg else len(pods)
+            if pg is None:
+                continue  # gang pods wait for their PodGroup to exist
+            min_member = pg.spec.min_member
             bound_members = [p for p in scheduled
                              if (p.metadata.annotations or {}).get(
                                  POD_GROUP_ANNOTATION) == gname]

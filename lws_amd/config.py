@@ -1,0 +1,123 @@
+"""Controller-manager configuration.
+
+Mirrors the reference component-config API
+(api/config/v1alpha1/configuration_types.go + pkg/config/): a strictly
+decoded YAML file with defaulting and validation, mapped onto the
+standalone manager's options.  Kubernetes-specific knobs (webhook certs,
+leader election) become their lws_amd equivalents: the API server bind
+address, node inventory, and the gang-scheduling provider.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+import yaml
+
+from .api import serde
+from .schedulerprovider.provider import SUPPORTED_PROVIDERS
+
+
+@dataclass
+class HealthEndpoints:
+    health_probe_bind_address: str = ":8081"
+
+
+@dataclass
+class MetricsConfig:
+    bind_address: str = ":8443"
+
+
+@dataclass
+class ApiServerConfig:
+    bind_address: str = "127.0.0.1:8080"
+    enable: bool = True
+
+
+@dataclass
+class GangSchedulingConfig:
+    # reference: Configuration.GangScheduling.SchedulerProvider
+    scheduler_provider: str = ""
+
+
+@dataclass
+class ClientConnectionConfig:
+    qps: float = 500.0
+    burst: int = 500
+
+
+@dataclass
+class NodeConfig:
+    name: str = ""
+    gpus: int = 8
+    labels: dict[str, str] = field(default_factory=dict)
+    address: str = "127.0.0.1"
+
+
+@dataclass
+class Configuration:
+    api_version: str = "config.lws.amd.com/v1alpha1"
+    kind: str = "Configuration"
+    health: HealthEndpoints = field(default_factory=HealthEndpoints)
+    metrics: MetricsConfig = field(default_factory=MetricsConfig)
+    api_server: ApiServerConfig = field(default_factory=ApiServerConfig)
+    gang_scheduling: GangSchedulingConfig = field(
+        default_factory=GangSchedulingConfig)
+    client_connection: ClientConnectionConfig = field(
+        default_factory=ClientConnectionConfig)
+    nodes: list[NodeConfig] = field(default_factory=list)
+    topology_key: str = "topology.lws.amd.com/island"
+
+
+class ConfigError(ValueError):
+    pass
+
+
+def _check_unknown_keys(data: dict, cls, path: str = "") -> None:
+    """Strict decode (pkg/config/config.go:35-46 uses strict yaml)."""
+    import dataclasses
+
+    known = {f.metadata.get("json", serde.snake_to_camel(f.name)): f
+             for f in dataclasses.fields(cls)}
+    for key, value in (data or {}).items():
+        if key not in known:
+            raise ConfigError(f"unknown field {path}{key!r}")
+        f = known[key]
+        ftype = f.type
+        import typing
+        hints = typing.get_type_hints(cls)
+        ft = hints[f.name]
+        if dataclasses.is_dataclass(ft) and isinstance(value, dict):
+            _check_unknown_keys(value, ft, path=f"{path}{key}.")
+
+
+def load(path: str) -> Configuration:
+    with open(path) as f:
+        data = yaml.safe_load(f) or {}
+    return from_dict(data)
+
+
+def from_dict(data: dict) -> Configuration:
+    _check_unknown_keys(data, Configuration)
+    for i, node in enumerate(data.get("nodes", []) or []):
+        _check_unknown_keys(node, NodeConfig, path=f"nodes[{i}].")
+    cfg = serde.from_dict(Configuration, data)
+    validate(cfg)
+    return cfg
+
+
+def validate(cfg: Configuration) -> None:
+    """pkg/config/validation.go equivalent."""
+    sp = cfg.gang_scheduling.scheduler_provider
+    if sp and sp not in SUPPORTED_PROVIDERS:
+        raise ConfigError(
+            f"gangScheduling.schedulerProvider {sp!r} not in "
+            f"{SUPPORTED_PROVIDERS}")
+    if cfg.client_connection.qps <= 0 or cfg.client_connection.burst <= 0:
+        raise ConfigError("clientConnection qps/burst must be positive")
+    names = [n.name for n in cfg.nodes]
+    if len(set(names)) != len(names):
+        raise ConfigError("node names must be unique")
+    for n in cfg.nodes:
+        if n.gpus < 0:
+            raise ConfigError(f"node {n.name}: gpus must be >= 0")

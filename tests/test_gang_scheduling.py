@@ -1,0 +1,103 @@
+"""Gang scheduling tests (mirrors reference e2e_gang_scheduling_test.go):
+PodGroup creation with MinMember/MinResources, all-or-nothing binding,
+no partial scheduling when capacity is insufficient."""
+import time
+
+import pytest
+
+from lws_amd.api import leaderworkerset as lwsapi
+from tests.conftest import lws_condition, make_lws, wait_for
+
+
+def _gang_cluster(nodes):
+    from lws_amd.cluster.cluster import LwsCluster
+    from lws_amd.schedulerprovider.provider import GangProvider
+
+    return LwsCluster(nodes=nodes,
+                      scheduler_provider_factory=GangProvider).start()
+
+
+def _gpu_lws(name, replicas, size, gpus=1):
+    lws = make_lws(name=name, replicas=replicas, size=size)
+    lws.spec.leader_worker_template.worker_template.spec.containers[0] \
+        .resources.requests = {"amd.com/gpu": gpus}
+    return lws
+
+
+def test_podgroup_created_and_gang_binds():
+    from lws_amd.cluster.cluster import make_nodes
+
+    c = _gang_cluster(make_nodes(1, gpus_per_node=8))
+    try:
+        c.store.create(_gpu_lws("gang", replicas=1, size=4))
+
+        pgs = wait_for(lambda: c.store.list("PodGroup", "default") or None,
+                       desc="PodGroup", timeout=20)
+        assert len(pgs) == 1
+        pg = pgs[0]
+        assert pg.spec.min_member == 4
+        assert int(pg.spec.min_resources["amd.com/gpu"]) == 4
+        assert pg.metadata.name.startswith("gang-0-")
+
+        def available():
+            cur = c.get_lws("default", "gang")
+            cond = lws_condition(cur, "Available")
+            return cur if cond is not None and cond.status == "True" else None
+        wait_for(available, desc="gang group Available", timeout=30)
+        pods = c.store.list("Pod", "default")
+        assert len(pods) == 4 and all(p.node_name for p in pods)
+        # pods carry the gang annotation injected by the webhook
+        for p in pods:
+            assert p.metadata.annotations[
+                "scheduling.k8s.io/group-name"] == pg.metadata.name
+    finally:
+        c.stop()
+
+
+def test_gang_no_partial_binding_on_insufficient_capacity():
+    from lws_amd.cluster.cluster import make_nodes
+
+    # 2 GPUs total but the gang needs 4 -> nothing may bind
+    c = _gang_cluster(make_nodes(1, gpus_per_node=2))
+    try:
+        c.store.create(_gpu_lws("toolarge", replicas=1, size=4))
+        wait_for(lambda: len(c.store.list("Pod", "default")) == 4,
+                 desc="4 pending pods", timeout=20)
+        time.sleep(0.5)
+        pods = c.store.list("Pod", "default")
+        assert all(not p.node_name for p in pods), \
+            "gang must not partially bind"
+    finally:
+        c.stop()
+
+
+def test_gang_binds_after_capacity_frees():
+    from lws_amd.cluster.cluster import make_nodes
+
+    c = _gang_cluster(make_nodes(1, gpus_per_node=4))
+    try:
+        c.store.create(_gpu_lws("first", replicas=1, size=4))
+        wait_for(lambda: all(p.node_name for p in
+                             c.store.list("Pod", "default")) and
+                 len(c.store.list("Pod", "default")) == 4,
+                 desc="first gang bound", timeout=20)
+
+        c.store.create(_gpu_lws("second", replicas=1, size=4))
+        time.sleep(0.4)
+        second = [p for p in c.store.list("Pod", "default")
+                  if p.metadata.name.startswith("second")]
+        assert all(not p.node_name for p in second), "second gang must queue"
+
+        c.store.delete(lwsapi.KIND, "default", "first",
+                       propagation="Background")
+
+        def second_bound():
+            pods = [p for p in c.store.list("Pod", "default")
+                    if p.metadata.name.startswith("second")]
+            if len(pods) != 4 or any(not p.node_name for p in pods):
+                return None
+            return pods
+        wait_for(second_bound, desc="second gang bound after free",
+                 timeout=30)
+    finally:
+        c.stop()

@@ -1,0 +1,126 @@
+"""Manager + API server + client e2e (the kind-cluster e2e analogue):
+boot `python -m lws_amd`, drive it with the typed client and lwsctl."""
+import os
+import random
+import subprocess
+import sys
+import time
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture
+def manager_proc():
+    port = random.randint(20000, 60000)
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "lws_amd", "--api-bind", f"127.0.0.1:{port}",
+         "--nodes", "1", "--scheduler-provider", "gang"],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True)
+    base = f"http://127.0.0.1:{port}"
+    from lws_amd.client.clientset import HttpTransport
+
+    t = HttpTransport(base)
+    deadline = time.monotonic() + 60
+    while time.monotonic() < deadline:
+        if proc.poll() is not None:
+            raise AssertionError(f"manager died: {proc.stderr.read()[-2000:]}")
+        if t.healthz():
+            break
+        time.sleep(0.1)
+    else:
+        proc.kill()
+        raise AssertionError("manager never became healthy")
+    yield proc, base
+    proc.terminate()
+    try:
+        proc.wait(timeout=10)
+    except subprocess.TimeoutExpired:
+        proc.kill()
+
+
+def test_manager_api_and_client(manager_proc):
+    proc, base = manager_proc
+    from lws_amd.client.clientset import Clientset
+    from tests.conftest import make_lws, wait_for
+
+    cs = Clientset.for_server(base)
+    lws_client = cs.leader_worker_sets("default")
+    created = lws_client.create(make_lws(name="api-lws", replicas=1, size=2))
+    assert created.spec.rollout_strategy.type == "RollingUpdate"  # defaulted
+
+    def ready():
+        cur = lws_client.get("api-lws")
+        return cur if cur and cur.status.ready_replicas == 1 else None
+    wait_for(ready, desc="ready via API", timeout=30)
+
+    pods = cs.pods("default").list()
+    assert len(pods) == 2
+    env = {e.name: e.value for e in pods[0].spec.containers[0].env}
+    assert "LWS_LEADER_ADDRESS" in env
+
+    # scale subresource (HPA path)
+    lws_client.scale("api-lws", 2)
+    wait_for(lambda: len(cs.pods("default").list()) == 4,
+             desc="scaled to 4 pods", timeout=30)
+    scale = lws_client.get_scale("api-lws")
+    assert scale["spec"]["replicas"] == 2
+    assert "worker-index=0" in scale["status"]["selector"]
+
+    # metrics endpoint
+    import httpx
+    m = httpx.get(f"{base}/metrics").text
+    assert 'lws_amd_objects{kind="Pod"}' in m
+
+    lws_client.delete("api-lws")
+    wait_for(lambda: not cs.pods("default").list(), desc="pods gone",
+             timeout=30)
+
+
+def test_lwsctl_cli(manager_proc):
+    proc, base = manager_proc
+
+    def ctl(*args):
+        r = subprocess.run(
+            [sys.executable, "-m", "lws_amd.client.ctl", "--server", base,
+             *args], capture_output=True, text=True, cwd=REPO, timeout=60)
+        assert r.returncode == 0, r.stderr
+        return r.stdout
+
+    out = ctl("apply", "-f", os.path.join(REPO, "examples/lws-basic.yaml"))
+    assert "applied" in out
+    deadline = time.monotonic() + 30
+    while time.monotonic() < deadline:
+        out = ctl("get", "lws")
+        if "my-lws" in out:
+            break
+        time.sleep(0.2)
+    out = ctl("get", "lws", "my-lws", "-o", "yaml")
+    assert "leaderWorkerTemplate" in out
+    ctl("scale", "lws", "my-lws", "--replicas", "1")
+    ctl("delete", "lws", "my-lws")
+    out = ctl("get", "lws")
+    assert "my-lws" not in out
+
+
+def test_informer(manager_proc):
+    proc, base = manager_proc
+    from lws_amd.client.clientset import Clientset, Informer
+    from tests.conftest import make_lws, wait_for
+
+    cs = Clientset.for_server(base)
+    rc = cs.leader_worker_sets("default")
+    events = []
+    inf = Informer(rc, resync_seconds=0.2).start()
+    inf.add_handler(lambda ev, obj: events.append((ev, obj.metadata.name)))
+    try:
+        rc.create(make_lws(name="watched", replicas=1, size=1))
+        wait_for(lambda: ("ADDED", "watched") in events, desc="ADDED event",
+                 timeout=20)
+        assert any(o.metadata.name == "watched" for o in inf.lister())
+        rc.delete("watched")
+        wait_for(lambda: ("DELETED", "watched") in events,
+                 desc="DELETED event", timeout=20)
+    finally:
+        inf.stop()
