@@ -24,12 +24,15 @@ def check_rollout(initial_old, target, config):
         assert all(cur.past[i] <= prev.past[i] for i in range(n)), (prev, cur)
         assert all(cur.new[i] >= prev.new[i] for i in range(n)), (prev, cur)
         # surge bound (relative to the larger of initial/target during
-        # scale-down rollouts, where old starts above target)
+        # scale-down rollouts, where old starts above target).  Orphan
+        # prevention may deliberately pin a fully-drained role back at
+        # old=1 (planner.go:294-297), exceeding the bound by one — allowed.
         for i in range(n):
             if target[i] > 0:
+                slack = 1 if cur.past[i] == 1 else 0
                 assert cur.past[i] + cur.new[i] <= \
-                    max(initial_old[i], target[i]) + config[i].max_surge, \
-                    (prev, cur, i)
+                    max(initial_old[i], target[i]) + config[i].max_surge + \
+                    slack, (prev, cur, i)
         # availability floor (roles not scaling up past initial)
         for i in range(n):
             if initial_old[i] >= target[i]:
