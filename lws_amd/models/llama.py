@@ -35,6 +35,9 @@ class LlamaConfig:
     rope_theta: float = 500000.0
     rms_eps: float = 1e-5
     max_position: int = 8192
+    # MoE (Mixtral-style); num_experts == 0 means dense SwiGLU MLP
+    num_experts: int = 0
+    num_experts_per_tok: int = 2
 
     @property
     def q_size(self) -> int:
@@ -46,8 +49,11 @@ class LlamaConfig:
 
     def num_params(self) -> int:
         h, i, v = self.hidden_size, self.intermediate_size, self.vocab_size
+        mlp = 3 * h * i
+        if self.num_experts > 0:
+            mlp = self.num_experts * 3 * h * i + h * self.num_experts
         per_layer = (h * self.q_size + 2 * h * self.kv_size + self.q_size * h
-                     + 3 * h * i + 2 * h)
+                     + mlp + 2 * h)
         return self.num_layers * per_layer + 2 * v * h + h
 
 
@@ -63,6 +69,22 @@ def llama3_70b() -> LlamaConfig:
                        num_kv_heads=8, vocab_size=128256)
 
 
+def mixtral_8x7b() -> LlamaConfig:
+    """Mixtral-8x7B (public architecture dims): 8 experts, top-2 routing."""
+    return LlamaConfig(name="mixtral-8x7b", hidden_size=4096,
+                       intermediate_size=14336, num_layers=32, num_q_heads=32,
+                       num_kv_heads=8, vocab_size=32000, rope_theta=1e6,
+                       num_experts=8, num_experts_per_tok=2)
+
+
+def mixtral_tiny() -> LlamaConfig:
+    return LlamaConfig(name="mixtral-tiny", hidden_size=256,
+                       intermediate_size=512, num_layers=2, num_q_heads=8,
+                       num_kv_heads=2, head_dim=128, vocab_size=1024,
+                       max_position=2048, num_experts=4,
+                       num_experts_per_tok=2)
+
+
 def llama_tiny() -> LlamaConfig:
     """Small config for tests/smoke (structure-identical to Llama-3)."""
     return LlamaConfig(name="llama-tiny", hidden_size=256,
@@ -75,6 +97,8 @@ MODEL_PRESETS = {
     "llama-3-8b": llama3_8b,
     "llama-3-70b": llama3_70b,
     "llama-tiny": llama_tiny,
+    "mixtral-8x7b": mixtral_8x7b,
+    "mixtral-tiny": mixtral_tiny,
 }
 
 
@@ -131,20 +155,56 @@ class LlamaLayer:
         self.qkv = ShardedLinear(cfg.q_size + 2 * cfg.kv_size, h, 0,
                                  tp_rank, tp_world, device, dtype)
         self.o = ShardedLinear(h, cfg.q_size, 1, tp_rank, tp_world, device, dtype)
-        self.gate_up = ShardedLinear(2 * cfg.intermediate_size, h, 0,
-                                     tp_rank, tp_world, device, dtype)
-        self.down = ShardedLinear(h, cfg.intermediate_size, 1, tp_rank,
-                                  tp_world, device, dtype)
+        if cfg.num_experts > 0:
+            # Mixtral MoE: router replicated; every expert's SwiGLU is
+            # TP-sharded like the dense MLP (intra-node tensor parallel over
+            # xGMI; expert-parallel across groups is the DS role split)
+            self.router = torch.empty(cfg.num_experts, h, device=device,
+                                      dtype=dtype)
+            self.experts_gate_up = [
+                ShardedLinear(2 * cfg.intermediate_size, h, 0, tp_rank,
+                              tp_world, device, dtype)
+                for _ in range(cfg.num_experts)]
+            self.experts_down = [
+                ShardedLinear(h, cfg.intermediate_size, 1, tp_rank, tp_world,
+                              device, dtype)
+                for _ in range(cfg.num_experts)]
+            self.gate_up = None
+            self.down = None
+        else:
+            self.router = None
+            self.gate_up = ShardedLinear(2 * cfg.intermediate_size, h, 0,
+                                         tp_rank, tp_world, device, dtype)
+            self.down = ShardedLinear(h, cfg.intermediate_size, 1, tp_rank,
+                                      tp_world, device, dtype)
         self.input_norm = torch.empty(h, device=device, dtype=dtype)
         self.post_norm = torch.empty(h, device=device, dtype=dtype)
         self.q_slice = self.hq * d
         self.kv_slice = self.hkv * d
 
     def materialize(self, gen=None):
-        for lin in (self.qkv, self.o, self.gate_up, self.down):
+        lins = [self.qkv, self.o]
+        if self.router is not None:
+            self.router.normal_(0.0, 0.02, generator=gen)
+            lins += self.experts_gate_up + self.experts_down
+        else:
+            lins += [self.gate_up, self.down]
+        for lin in lins:
             lin.materialize(gen)
         self.input_norm.fill_(1.0)
         self.post_norm.fill_(1.0)
+
+    @property
+    def param_count(self) -> int:
+        n = self.qkv.numel + self.o.numel
+        n += self.input_norm.numel() + self.post_norm.numel()
+        if self.router is not None:
+            n += self.router.numel()
+            n += sum(l.numel for l in self.experts_gate_up)
+            n += sum(l.numel for l in self.experts_down)
+        else:
+            n += self.gate_up.numel + self.down.numel
+        return n
 
 
 class LlamaForCausalLM:
@@ -177,9 +237,7 @@ class LlamaForCausalLM:
         n = 0
         for layer in self.layers:
             layer.materialize(gen)
-            n += sum(l.numel for l in (layer.qkv, layer.o, layer.gate_up,
-                                       layer.down))
-            n += layer.input_norm.numel() + layer.post_norm.numel()
+            n += layer.param_count
         self.embed.normal_(0.0, 0.02, generator=gen)
         self.final_norm.fill_(1.0)
         self.lm_head.materialize(gen)
@@ -274,9 +332,7 @@ class LlamaForCausalLM:
             x = all_reduce(o)
             # MLP
             h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)
-            gu = layer.gate_up(h)
-            act = self._silu_mul(gu)
-            x = all_reduce(layer.down(act))
+            x = all_reduce(self._mlp(layer, h))
         h, _ = self._fused_add_rmsnorm(x, residual, self.final_norm)
         last = torch.tensor([s - 1 for s in batch.seq_starts[1:]],
                             device=h.device, dtype=torch.long)
@@ -335,11 +391,32 @@ class LlamaForCausalLM:
             o = layer.o(attn.view(B, layer.hq * cfg.head_dim))
             x = all_reduce(o)
             h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)
-            gu = layer.gate_up(h)
-            act = self._silu_mul(gu)
-            x = all_reduce(layer.down(act))
+            x = all_reduce(self._mlp(layer, h))
         h, _ = self._fused_add_rmsnorm(x, residual, self.final_norm)
         return h
+
+    def _mlp(self, layer, h: torch.Tensor) -> torch.Tensor:
+        """Dense SwiGLU or Mixtral top-k MoE (router in fp32; each chosen
+        expert's TP-sharded SwiGLU applied to its token subset)."""
+        if layer.router is None:
+            return layer.down(self._silu_mul(layer.gate_up(h)))
+        cfg = self.cfg
+        logits = (h.float() @ layer.router.float().t())      # [T, E]
+        weights, chosen = torch.topk(torch.softmax(logits, dim=-1),
+                                     cfg.num_experts_per_tok, dim=-1)
+        weights = weights / weights.sum(dim=-1, keepdim=True)
+        out = torch.zeros_like(h)
+        for e in range(cfg.num_experts):
+            mask = (chosen == e)
+            token_idx, slot_idx = mask.nonzero(as_tuple=True)
+            if token_idx.numel() == 0:
+                continue
+            xe = h[token_idx]
+            ye = layer.experts_down[e](
+                self._silu_mul(layer.experts_gate_up[e](xe)))
+            w = weights[token_idx, slot_idx].unsqueeze(-1).to(ye.dtype)
+            out.index_add_(0, token_idx, ye * w)
+        return out
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         """[N, H] -> [N, vocab] (column-sharded lm_head + all-gather)."""

@@ -1,0 +1,100 @@
+"""Served-container entrypoint: `python -m lws_amd.serving.launch`.
+
+This is what runs inside each group pod of a real deployment (the
+reference's vLLM-container analogue — SURVEY.md §2.9).  It consumes the
+rendezvous env the lws_amd pod webhook injects:
+
+    LWS_LEADER_ADDRESS / LWS_GROUP_SIZE / LWS_WORKER_INDEX  (identity)
+    MASTER_ADDR / MASTER_PORT / WORLD_SIZE / NODE_RANK /
+    LOCAL_WORLD_SIZE                                         (RCCL)
+
+Each process hosts one TP shard on one MI355X.  Worker index 0 (the
+leader) additionally serves the OpenAI-style HTTP frontend and broadcasts
+engine commands; workers follow (same collective protocol as bench.py).
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(prog="lws-amd-engine")
+    p.add_argument("--model", default=os.environ.get("LWS_AMD_MODEL",
+                                                     "llama-tiny"))
+    p.add_argument("--kv-pages", type=int,
+                   default=int(os.environ.get("LWS_AMD_KV_PAGES", "512")))
+    p.add_argument("--port", type=int,
+                   default=int(os.environ.get("LWS_AMD_HTTP_PORT", "8000")))
+    p.add_argument("--device", default=None)
+    p.add_argument("--seed", type=int, default=0)
+    args = p.parse_args(argv)
+
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE",
+                               os.environ.get("LWS_GROUP_SIZE", "1")))
+    rank = int(os.environ.get("RANK",
+                              os.environ.get("NODE_RANK",
+                                             os.environ.get("LWS_WORKER_INDEX",
+                                                            "0"))))
+    os.environ.setdefault("RANK", str(rank))
+    os.environ.setdefault("WORLD_SIZE", str(world))
+    if "MASTER_ADDR" not in os.environ and "LWS_LEADER_ADDRESS" in os.environ:
+        os.environ["MASTER_ADDR"] = os.environ["LWS_LEADER_ADDRESS"]
+
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    if device.startswith("cuda"):
+        import lws_amd.ops as ops
+        ops.require_native()  # no silent eager fallback on a GPU host
+
+    from lws_amd.parallel.tp import init_distributed
+    if world > 1:
+        init_distributed(device=None if device == "cuda" else device)
+
+    from lws_amd.serving.engine import Engine, EngineConfig
+    from lws_amd.serving.runtime import Conductor, WorkerLoop
+
+    import torch.distributed as dist
+    control = dist.new_group(backend="gloo") if world > 1 else None
+
+    if rank != 0:
+        WorkerLoop(rank, world, device, control).run()
+        return 0
+
+    conductor = Conductor(0, world, device, control)
+    conductor.command({"op": "build",
+                       "spec": {"model": args.model,
+                                "kv_pages": args.kv_pages,
+                                "seed": args.seed}})
+    engine = conductor.host.engine
+
+    from lws_amd.serving.server import ServingLoop, build_app
+
+    # NOTE: with world > 1 the engine steps are collective; the leader's
+    # serving loop must broadcast step commands.  v1 scope: HTTP serving is
+    # single-shard (TP=1) or driven externally via bench.py's conductor.
+    if world > 1:
+        print("lws-amd-engine: leader ready (collective mode); "
+              "serving frontend requires the conductor protocol",
+              flush=True)
+        try:
+            import time
+            while True:
+                time.sleep(1)
+        except KeyboardInterrupt:
+            conductor.command({"op": "exit"})
+        return 0
+
+    loop = ServingLoop(engine).start()
+    app = build_app(loop, args.model)
+    import uvicorn
+
+    print(f"lws-amd-engine: serving {args.model} on :{args.port}", flush=True)
+    uvicorn.run(app, host="0.0.0.0", port=args.port, log_level="warning")
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

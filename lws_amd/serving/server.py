@@ -1,0 +1,153 @@
+"""OpenAI-style HTTP frontend for the lws_amd serving engine.
+
+Runs on the group leader (worker index 0); TP shard workers join the same
+RCCL process group and follow broadcast commands.  Endpoints:
+
+    POST /v1/completions   {prompt: [ids] | str, max_tokens, temperature}
+    GET  /v1/models
+    GET  /health           (the readiness signal the node agent probes)
+    GET  /metrics          (Prometheus-style serving counters)
+
+Requests queue into the engine's continuous-batching loop; a background
+stepper thread drives prefill/decode steps and resolves futures as
+sequences finish.
+"""
+from __future__ import annotations
+
+import threading
+import time
+import uuid
+from concurrent.futures import Future
+from dataclasses import dataclass, field
+from typing import Optional
+
+from .engine import Engine, EngineConfig
+
+
+@dataclass
+class _Pending:
+    seq_id: int
+    prompt_len: int
+    max_tokens: int
+    future: Future = field(default_factory=Future)
+    created: float = field(default_factory=time.time)
+
+
+class ServingLoop:
+    """Continuous-batching driver around Engine.step()."""
+
+    def __init__(self, engine: Engine):
+        self.engine = engine
+        self._lock = threading.Lock()
+        self._pending: dict[int, _Pending] = {}
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self.stats = {"requests": 0, "tokens_generated": 0, "steps": 0}
+
+    def start(self) -> "ServingLoop":
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+        return self
+
+    def submit(self, prompt_ids: list[int], max_tokens: int) -> Future:
+        with self._lock:
+            sid = self.engine.add_request(list(prompt_ids))
+            p = _Pending(seq_id=sid, prompt_len=len(prompt_ids),
+                         max_tokens=max_tokens)
+            self._pending[sid] = p
+            self.stats["requests"] += 1
+        return p.future
+
+    def _run(self) -> None:
+        while not self._stop.is_set():
+            with self._lock:
+                if not self._pending:
+                    pass
+                else:
+                    out = self.engine.step()
+                    self.stats["steps"] += 1
+                    self.stats["tokens_generated"] += len(out)
+                    finished = []
+                    for sid in list(self._pending):
+                        p = self._pending[sid]
+                        seq = self.engine.sequences.get(sid)
+                        if seq is None:
+                            finished.append(sid)
+                            continue
+                        produced = len(seq.token_ids) - p.prompt_len
+                        if produced >= p.max_tokens:
+                            tokens = seq.token_ids[
+                                p.prompt_len:p.prompt_len + p.max_tokens]
+                            self.engine.finish(sid)
+                            p.future.set_result(tokens)
+                            finished.append(sid)
+                    for sid in finished:
+                        self._pending.pop(sid, None)
+                    continue
+            time.sleep(0.002)
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=5)
+
+
+def build_app(loop: ServingLoop, model_name: str):
+    from fastapi import FastAPI, HTTPException
+
+    app = FastAPI(title="lws-amd-engine")
+
+    @app.get("/health")
+    def health():
+        if not loop.engine.ready:
+            raise HTTPException(503, "engine not ready")
+        return {"status": "ok"}
+
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list",
+                "data": [{"id": model_name, "object": "model",
+                          "owned_by": "lws-amd"}]}
+
+    @app.get("/metrics")
+    def metrics():
+        from fastapi.responses import PlainTextResponse
+
+        s = loop.stats
+        body = "\n".join([
+            "# TYPE lws_amd_engine_requests_total counter",
+            f"lws_amd_engine_requests_total {s['requests']}",
+            "# TYPE lws_amd_engine_tokens_generated_total counter",
+            f"lws_amd_engine_tokens_generated_total {s['tokens_generated']}",
+            "# TYPE lws_amd_engine_steps_total counter",
+            f"lws_amd_engine_steps_total {s['steps']}",
+        ]) + "\n"
+        return PlainTextResponse(body)
+
+    @app.post("/v1/completions")
+    def completions(body: dict):
+        prompt = body.get("prompt", [])
+        if isinstance(prompt, str):
+            # no tokenizer shipped (offline, random-init weights): encode
+            # bytes as token ids modulo vocab
+            vocab = loop.engine.model_cfg.vocab_size
+            prompt = [b % vocab for b in prompt.encode()] or [0]
+        max_tokens = int(body.get("max_tokens", 16))
+        fut = loop.submit(prompt, max_tokens)
+        tokens = fut.result(timeout=float(body.get("timeout", 300)))
+        return {
+            "id": f"cmpl-{uuid.uuid4().hex[:12]}",
+            "object": "text_completion",
+            "model": model_name,
+            "choices": [{
+                "index": 0,
+                "token_ids": tokens,
+                "text": " ".join(str(t) for t in tokens),
+                "finish_reason": "length",
+            }],
+            "usage": {"prompt_tokens": len(prompt),
+                      "completion_tokens": len(tokens),
+                      "total_tokens": len(prompt) + len(tokens)},
+        }
+
+    return app

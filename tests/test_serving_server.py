@@ -1,0 +1,64 @@
+"""Serving frontend tests: continuous batching loop + OpenAI-style API."""
+import threading
+import time
+
+import pytest
+
+
+def _make_loop():
+    from lws_amd.serving.engine import Engine, EngineConfig
+    from lws_amd.serving.server import ServingLoop
+
+    eng = Engine(EngineConfig(model="llama-tiny", kv_pages=64, device="cpu"))
+    eng.load()
+    return ServingLoop(eng).start()
+
+
+def test_serving_loop_completes_requests():
+    loop = _make_loop()
+    try:
+        f1 = loop.submit([1, 2, 3], max_tokens=4)
+        f2 = loop.submit([9, 8, 7, 6], max_tokens=2)
+        t1 = f1.result(timeout=60)
+        t2 = f2.result(timeout=60)
+        assert len(t1) == 4 and len(t2) == 2
+        assert loop.stats["requests"] == 2
+    finally:
+        loop.stop()
+
+
+def test_serving_loop_concurrent_submit():
+    loop = _make_loop()
+    try:
+        futs = [loop.submit([i, i + 1], max_tokens=3) for i in range(6)]
+        outs = [f.result(timeout=120) for f in futs]
+        assert all(len(o) == 3 for o in outs)
+    finally:
+        loop.stop()
+
+
+def test_completions_endpoint():
+    from fastapi.testclient import TestClient
+    from lws_amd.serving.server import build_app
+
+    loop = _make_loop()
+    try:
+        app = build_app(loop, "llama-tiny")
+        client = TestClient(app)
+        r = client.get("/health")
+        assert r.status_code == 200
+        r = client.get("/v1/models")
+        assert r.json()["data"][0]["id"] == "llama-tiny"
+        r = client.post("/v1/completions",
+                        json={"prompt": [5, 6, 7], "max_tokens": 3})
+        assert r.status_code == 200
+        body = r.json()
+        assert len(body["choices"][0]["token_ids"]) == 3
+        assert body["usage"]["total_tokens"] == 6
+        r = client.post("/v1/completions",
+                        json={"prompt": "hello world", "max_tokens": 2})
+        assert r.status_code == 200
+        r = client.get("/metrics")
+        assert "lws_amd_engine_requests_total 2" in r.text
+    finally:
+        loop.stop()
