@@ -1,0 +1,165 @@
+"""Webhook + placement behaviors: subgroups, UniquePerReplica subdomains,
+RecreateGroupAfterStart gate, RCCL env injection for GPU pods.
+(Mirrors reference pod_webhook_test.go / e2e subgroup + TPU-env scenarios.)
+"""
+import time
+
+import pytest
+
+from lws_amd.api import leaderworkerset as lwsapi
+from tests.conftest import lws_condition, make_lws, wait_for
+
+
+def _wait_available(cluster, name, timeout=30):
+    def available():
+        cur = cluster.get_lws("default", name)
+        cond = lws_condition(cur, "Available")
+        return cur if cond is not None and cond.status == "True" else None
+    return wait_for(available, desc=f"{name} Available", timeout=timeout)
+
+
+def test_subgroup_labels_leaderworker(cluster):
+    from lws_amd.api.leaderworkerset import SubGroupPolicy
+
+    lws = make_lws(name="sg", replicas=1, size=4)
+    lws.spec.leader_worker_template.sub_group_policy = SubGroupPolicy(
+        sub_group_size=2)
+    cluster.store.create(lws)
+    _wait_available(cluster, "sg")
+    pods = {p.metadata.name: p for p in cluster.store.list("Pod", "default")}
+    assert len(pods) == 4
+    # LeaderWorker: leader in subgroup 0; size=4, sgs=2 -> groups {0,1},{2,3}
+    assert pods["sg-0"].metadata.labels[lwsapi.SUBGROUP_INDEX_LABEL_KEY] == "0"
+    assert pods["sg-0-1"].metadata.labels[lwsapi.SUBGROUP_INDEX_LABEL_KEY] == "0"
+    assert pods["sg-0-2"].metadata.labels[lwsapi.SUBGROUP_INDEX_LABEL_KEY] == "1"
+    assert pods["sg-0-3"].metadata.labels[lwsapi.SUBGROUP_INDEX_LABEL_KEY] == "1"
+    # same subgroup -> same subgroup key
+    k01 = pods["sg-0-1"].metadata.labels[lwsapi.SUBGROUP_UNIQUE_HASH_LABEL_KEY]
+    k23 = pods["sg-0-2"].metadata.labels[lwsapi.SUBGROUP_UNIQUE_HASH_LABEL_KEY]
+    assert k01 != k23
+    assert pods["sg-0"].metadata.labels[
+        lwsapi.SUBGROUP_UNIQUE_HASH_LABEL_KEY] == k01
+    # size annotation propagated
+    assert pods["sg-0-1"].metadata.annotations[
+        lwsapi.SUBGROUP_SIZE_ANNOTATION_KEY] == "2"
+
+
+def test_subgroup_validation(cluster):
+    from lws_amd.api.leaderworkerset import SubGroupPolicy
+    from lws_amd.cluster.store import InvalidError
+
+    bad = make_lws(name="sgbad", size=4)
+    bad.spec.leader_worker_template.sub_group_policy = SubGroupPolicy(
+        sub_group_size=5)
+    with pytest.raises(InvalidError):
+        cluster.store.create(bad)
+
+    # immutability on update
+    ok = make_lws(name="sgok", size=4)
+    ok.spec.leader_worker_template.sub_group_policy = SubGroupPolicy(
+        sub_group_size=2)
+    cluster.store.create(ok)
+    cur = cluster.get_lws("default", "sgok")
+    cur.spec.leader_worker_template.sub_group_policy.sub_group_size = 4
+    with pytest.raises(InvalidError):
+        cluster.store.update(cur)
+
+
+def test_unique_per_replica_subdomain(cluster):
+    from lws_amd.api.leaderworkerset import NetworkConfig
+
+    lws = make_lws(name="upr", replicas=2, size=2)
+    lws.spec.network_config = NetworkConfig(
+        subdomain_policy="UniquePerReplica")
+    cluster.store.create(lws)
+    _wait_available(cluster, "upr")
+    pods = {p.metadata.name: p for p in cluster.store.list("Pod", "default")}
+    # leader pod subdomain == its own name; workers join it
+    assert pods["upr-0"].spec.subdomain == "upr-0"
+    assert pods["upr-0-1"].spec.subdomain == "upr-0"
+    assert pods["upr-1"].spec.subdomain == "upr-1"
+    env = {e.name: e.value
+           for e in pods["upr-0-1"].spec.containers[0].env}
+    assert env[lwsapi.LWS_LEADER_ADDRESS] == "upr-0.upr-0.default"
+    # per-replica headless services exist; no shared service
+    svcs = {s.metadata.name for s in cluster.store.list("Service", "default")}
+    assert {"upr-0", "upr-1"} <= svcs
+    assert "upr" not in svcs
+
+
+def test_rccl_env_injection_for_gpu_pods(cluster):
+    lws = make_lws(name="rccl", replicas=1, size=2)
+    lws.spec.leader_worker_template.worker_template.spec.containers[0] \
+        .resources.requests = {"amd.com/gpu": 4}
+    cluster.store.create(lws)
+    _wait_available(cluster, "rccl")
+    pods = {p.metadata.name: p for p in cluster.store.list("Pod", "default")}
+    env = {e.name: e.value for e in pods["rccl-0-1"].spec.containers[0].env}
+    assert env["MASTER_ADDR"] == "rccl-0.rccl.default"
+    assert env["WORLD_SIZE"] == "8"           # 2 pods x 4 GPUs
+    assert env["NODE_RANK"] == "1"
+    assert env["LOCAL_WORLD_SIZE"] == "4"
+    env0 = {e.name: e.value for e in pods["rccl-0"].spec.containers[0].env}
+    assert env0["NODE_RANK"] == "0"
+
+
+def test_recreate_group_after_start_gate():
+    """RecreateGroupAfterStart: restarts are ignored while any pod in the
+    group is still Pending (pod_controller.go:215-225)."""
+    from lws_amd.cluster.cluster import LwsCluster, make_nodes
+    from lws_amd.cluster.node import FakeRuntime
+
+    # slow runtime so pods stay Pending for a while
+    c = LwsCluster(nodes=make_nodes(1, gpus_per_node=8),
+                   runtime_factory=lambda n: FakeRuntime(ready_delay=0.8)).start()
+    try:
+        lws = make_lws(name="ags", replicas=1, size=2)
+        lws.spec.leader_worker_template.restart_policy = \
+            "RecreateGroupAfterStart"
+        c.store.create(lws)
+        pods = wait_for(lambda: (lambda ps: ps if len(ps) == 2 else None)(
+            c.store.list("Pod", "default")), desc="2 pods", timeout=20)
+        uids = {p.metadata.name: p.metadata.uid for p in pods}
+        # pods still Pending: simulate a restart -> group must NOT recreate
+        from lws_amd.api.core import ContainerStatus
+        from lws_amd.cluster.store import ConflictError
+
+        def bump_restart(name, phase="Running"):
+            for _ in range(50):
+                cur = c.store.get("Pod", "default", name)
+                cur.status.phase = phase
+                if cur.status.container_statuses:
+                    cur.status.container_statuses[0].restart_count += 1
+                    cur.status.container_statuses[0].ready = False
+                else:
+                    cur.status.container_statuses = [ContainerStatus(
+                        name="main", ready=False, restart_count=1)]
+                try:
+                    c.store.update_status(cur)
+                    return
+                except ConflictError:
+                    time.sleep(0.01)
+            raise AssertionError("could not update pod status")
+
+        bump_restart("ags-0-1")
+        time.sleep(0.4)
+        now = {p.metadata.name: p.metadata.uid
+               for p in c.store.list("Pod", "default")}
+        assert now == uids, "group must not recreate while pods pending"
+        _wait_available(c, "ags", timeout=30)
+
+        # once started, a restart recreates the group
+        bump_restart("ags-0-1")
+
+        def recreated():
+            ps = c.store.list("Pod", "default")
+            if len(ps) != 2:
+                return None
+            fresh = {p.metadata.name: p.metadata.uid for p in ps}
+            if set(fresh) != set(uids) or \
+                    any(fresh[n] == uids[n] for n in fresh):
+                return None
+            return ps
+        wait_for(recreated, desc="group recreated after start", timeout=30)
+    finally:
+        c.stop()
