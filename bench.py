@@ -253,10 +253,24 @@ def main() -> None:
         if is_gpu:
             torch.cuda.synchronize()
 
+    def _teardown_dist():
+        # rank 0 hosts the rendezvous TCPStore; tearing the process group
+        # down explicitly (after a final barrier) prevents the native gloo
+        # destructor from hanging when rank 0 exits first
+        if world > 1:
+            try:
+                dist.barrier()
+                dist.destroy_process_group()
+            except Exception:  # noqa: BLE001
+                pass
+
     if rank != 0:
         # workers follow commands; block timing handled by execute_command
         WorkerLoop(rank, world, args.device, control_group, sync).run()
-        return
+        _teardown_dist()
+        sys.stdout.flush()
+        sys.stderr.flush()
+        os._exit(0)
 
     # ---- rank 0 ----
     bc = BenchConductor(args, world, args.device, control_group, sync)
@@ -312,6 +326,10 @@ def main() -> None:
         print(json.dumps(out), flush=True)
     finally:
         bc.shutdown()
+        _teardown_dist()
+        sys.stdout.flush()
+        sys.stderr.flush()
+        os._exit(0)
 
 
 if __name__ == "__main__":

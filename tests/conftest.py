@@ -53,3 +53,19 @@ def lws_condition(lws, cond_type):
         if c.type == cond_type:
             return c
     return None
+
+
+def retry_update(store, kind, namespace, name, mutate, attempts=50):
+    """Conflict-retrying spec update (controllers bump resourceVersion via
+    status writes concurrently; mirrors client-go retry.RetryOnConflict)."""
+    import time as _time
+    from lws_amd.cluster.store import ConflictError
+
+    for _ in range(attempts):
+        obj = store.get(kind, namespace, name)
+        mutate(obj)
+        try:
+            return store.update(obj)
+        except ConflictError:
+            _time.sleep(0.01)
+    raise AssertionError(f"update of {kind} {name} kept conflicting")

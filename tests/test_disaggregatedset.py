@@ -109,11 +109,13 @@ def test_ds_lockstep_rolling_update(cluster):
     old_lws = {l.metadata.name for l in cluster.store.list(lwsapi.KIND,
                                                            "default")}
 
-    cur = cluster.store.get(dsapi.KIND, "default", "my-ds")
-    for role in cur.spec.roles:
-        role.spec.leader_worker_template.worker_template.spec.containers[0] \
-            .image = "engine:v2"
-    cluster.store.update(cur)
+    from tests.conftest import retry_update
+
+    def set_images(o):
+        for role in o.spec.roles:
+            role.spec.leader_worker_template.worker_template.spec \
+                .containers[0].image = "engine:v2"
+    retry_update(cluster.store, dsapi.KIND, "default", "my-ds", set_images)
 
     def rolled():
         c = ds_available(cluster)
@@ -141,9 +143,9 @@ def test_ds_role_scale(cluster):
     cluster.store.create(ds)
     wait_for(lambda: ds_available(cluster), desc="Available", timeout=60)
 
-    cur = cluster.store.get(dsapi.KIND, "default", "my-ds")
-    cur.spec.roles[1].spec.replicas = 4
-    cluster.store.update(cur)
+    from tests.conftest import retry_update
+    retry_update(cluster.store, dsapi.KIND, "default", "my-ds",
+                 lambda o: setattr(o.spec.roles[1].spec, "replicas", 4))
 
     def scaled():
         c = ds_available(cluster)
@@ -173,9 +175,9 @@ def test_ds_external_scaler(cluster):
     wait_for(lambda: ds_available(cluster), desc="Available", timeout=60)
 
     # external autoscaler writes spec.replicas (the /scale path)
-    scaler = cluster.store.get(dsapi.SCALER_KIND, "default", "my-ds-decode")
-    scaler.spec.replicas = 3
-    cluster.store.update(scaler)
+    from tests.conftest import retry_update
+    retry_update(cluster.store, dsapi.SCALER_KIND, "default", "my-ds-decode",
+                 lambda o: setattr(o.spec, "replicas", 3))
 
     def scaled():
         c = ds_available(cluster)
@@ -211,8 +213,9 @@ def test_ds_slices():
             assert rs.replicas == 2
 
         # slice scale-down removes slice-1 objects
-        cur.spec.slices = 1
-        c.store.update(cur)
+        from tests.conftest import retry_update
+        retry_update(c.store, dsapi.KIND, "default", "my-ds",
+                     lambda o: setattr(o.spec, "slices", 1))
         def one_slice():
             lws_list = c.store.list(lwsapi.KIND, "default")
             if len(lws_list) != 2:

@@ -139,15 +139,14 @@ def test_scale_up_and_down(cluster):
     wait_for(lambda: len(cluster.store.list("Pod", "default")) == 2,
              desc="initial 2 pods")
 
-    cur = cluster.get_lws("default", "scale")
-    cur.spec.replicas = 3
-    cluster.store.update(cur)
+    from tests.conftest import retry_update
+    retry_update(cluster.store, "LeaderWorkerSet", "default", "scale",
+                 lambda o: setattr(o.spec, "replicas", 3))
     wait_for(lambda: len(cluster.store.list("Pod", "default")) == 6,
              desc="scale up to 6 pods")
 
-    cur = cluster.get_lws("default", "scale")
-    cur.spec.replicas = 1
-    cluster.store.update(cur)
+    retry_update(cluster.store, "LeaderWorkerSet", "default", "scale",
+                 lambda o: setattr(o.spec, "replicas", 1))
     wait_for(lambda: len(cluster.store.list("Pod", "default")) == 2,
              desc="scale down to 2 pods", timeout=30)
     names = sorted(p.metadata.name for p in cluster.store.list("Pod", "default"))

@@ -82,10 +82,13 @@ def test_rolling_update_replaces_all_groups(cluster):
                    for p in cluster.store.list("Pod", "default")}
     assert len(pods_before) == 6
 
-    cur = cluster.get_lws("default", "roll")
-    cur.spec.leader_worker_template.worker_template.spec.containers[0].image = \
-        "engine:v2"
-    cluster.store.update(cur)
+    from tests.conftest import retry_update
+
+    def set_image(o, img="engine:v2"):
+        o.spec.leader_worker_template.worker_template.spec.containers[0] \
+            .image = img
+    retry_update(cluster.store, "LeaderWorkerSet", "default", "roll",
+                 set_image)
 
     # UpdateInProgress should appear
     def updating():
@@ -129,10 +132,13 @@ def test_rolling_update_with_max_surge(cluster):
     cluster.store.create(lws)
     _wait_available(cluster, "surge")
 
-    cur = cluster.get_lws("default", "surge")
-    cur.spec.leader_worker_template.worker_template.spec.containers[0].image = \
-        "engine:v2"
-    cluster.store.update(cur)
+    from tests.conftest import retry_update
+
+    def set_image(o):
+        o.spec.leader_worker_template.worker_template.spec.containers[0] \
+            .image = "engine:v2"
+    retry_update(cluster.store, "LeaderWorkerSet", "default", "surge",
+                 set_image)
 
     def done():
         c = cluster.get_lws("default", "surge")
@@ -155,11 +161,14 @@ def test_partition_blocks_lower_ordinals(cluster):
     cluster.store.create(lws)
     _wait_available(cluster, "part")
 
-    cur = cluster.get_lws("default", "part")
-    cur.spec.rollout_strategy.rolling_update_configuration.partition = 2
-    cur.spec.leader_worker_template.worker_template.spec.containers[0].image = \
-        "engine:v2"
-    cluster.store.update(cur)
+    from tests.conftest import retry_update
+
+    def start_partitioned(o):
+        o.spec.rollout_strategy.rolling_update_configuration.partition = 2
+        o.spec.leader_worker_template.worker_template.spec.containers[0] \
+            .image = "engine:v2"
+    retry_update(cluster.store, "LeaderWorkerSet", "default", "part",
+                 start_partitioned)
 
     # only group 2 updates; groups 0,1 stay on v1
     def partial():
@@ -175,9 +184,10 @@ def test_partition_blocks_lower_ordinals(cluster):
     assert by_name["part-2"].spec.containers[0].image == "engine:v2"
 
     # release the partition → full rollout
-    cur = cluster.get_lws("default", "part")
-    cur.spec.rollout_strategy.rolling_update_configuration.partition = 0
-    cluster.store.update(cur)
+    retry_update(
+        cluster.store, "LeaderWorkerSet", "default", "part",
+        lambda o: setattr(o.spec.rollout_strategy
+                          .rolling_update_configuration, "partition", 0))
 
     def done():
         c = cluster.get_lws("default", "part")
