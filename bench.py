@@ -324,12 +324,22 @@ def main() -> None:
             },
         }
         print(json.dumps(out), flush=True)
+        ok = True
+    except BaseException:
+        import traceback
+        traceback.print_exc()
+        ok = False
     finally:
-        bc.shutdown()
+        try:
+            bc.shutdown()
+        except Exception:  # noqa: BLE001
+            pass
         _teardown_dist()
         sys.stdout.flush()
         sys.stderr.flush()
-        os._exit(0)
+        # hard exit: the control-plane daemon threads and the gloo
+        # destructor must not block process death
+        os._exit(0 if ok else 1)
 
 
 if __name__ == "__main__":
