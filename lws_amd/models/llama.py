@@ -339,9 +339,12 @@ class LlamaForCausalLM:
         return h[last]
 
     def _prefill_attention(self, q, k, v, batch: PrefillBatch, layer):
-        """Per-sequence causal attention over [T, H, D] views (library-GEMM
-        composition in bf16 with fp32 softmax; the decode path is the hot
-        loop and uses the HIP paged-attention kernel)."""
+        """Causal varlen attention over [T, H, D] views.  GPU: the in-repo
+        flash-style HIP kernel (MFMA, online softmax, no S x S
+        materialization).  CPU: per-sequence fp32 reference composition."""
+        if self._ops.is_gpu:
+            return self._ops.ops.prefill_attention(
+                q, k, v, batch.seq_starts, self.scale)
         cfg = self.cfg
         d = cfg.head_dim
         T = q.size(0)

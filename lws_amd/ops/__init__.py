@@ -174,3 +174,30 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
         _SKINNY_WS[key] = ws
     lib.skinny_gemm(out, x, w, ws)
     return out
+
+
+def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                      seq_starts: list, scale: float,
+                      out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Flash-style causal varlen prefill attention.
+
+    q [T, Hq, D] / k,v [T, Hkv, D] (strided row views OK, D=128);
+    seq_starts: python list of B+1 prefix offsets.  Returns [T, Hq*D].
+    """
+    lib = require_native()
+    T, Hq, D = q.shape
+    device = q.device
+    tile_seq, tile_q0 = [], []
+    for i in range(len(seq_starts) - 1):
+        S = seq_starts[i + 1] - seq_starts[i]
+        for q0 in range(0, S, 64):
+            tile_seq.append(i)
+            tile_q0.append(q0)
+    if out is None:
+        out = torch.empty(T, Hq * D, dtype=q.dtype, device=device)
+    lib.prefill_attention(
+        out, q, k, v,
+        torch.tensor(tile_seq, dtype=torch.int32, device=device),
+        torch.tensor(tile_q0, dtype=torch.int32, device=device),
+        torch.tensor(seq_starts, dtype=torch.int32, device=device), scale)
+    return out

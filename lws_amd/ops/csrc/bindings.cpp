@@ -19,6 +19,10 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v,
                        torch::Tensor slot_mapping);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws);
+void prefill_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                       torch::Tensor v, torch::Tensor tile_seq,
+                       torch::Tensor tile_q0, torch::Tensor seq_starts,
+                       double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, gfx950)");
@@ -32,4 +36,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "scatter new k/v into the paged KV cache");
   m.def("skinny_gemm", &skinny_gemm,
         "split-K MFMA GEMM for decode-shape projections (M<=32)");
+  m.def("prefill_attention", &prefill_attention,
+        "flash-style causal varlen prefill attention (GQA, gfx950)");
 }

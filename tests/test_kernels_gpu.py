@@ -191,3 +191,60 @@ def test_skinny_gemm(M, N, K):
     out = ops.skinny_gemm(x, w)
     ref = (x.float() @ w.float().t()).to(torch.bfloat16)
     assert_close_bf16(out, ref, atol=3e-2, rtol=3e-2)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("Hq,Hkv,lens", [
+    (8, 2, [64]),                  # single tile
+    (8, 8, [5]),                   # sub-tile MHA
+    (8, 1, [200, 64, 33, 128]),    # varlen batch, GQA 8
+    (16, 2, [513]),                # long, crosses many tiles
+])
+def test_prefill_attention(Hq, Hkv, lens):
+    import lws_amd.ops as ops
+    from lws_amd.ops.reference import sdpa_prefill_ref
+
+    D = 128
+    scale = 1.0 / math.sqrt(D)
+    T = sum(lens)
+    q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device="cuda")
+    starts = [0]
+    for L in lens:
+        starts.append(starts[-1] + L)
+    out = ops.prefill_attention(q, k, v, starts, scale)
+    # reference per sequence
+    for i, L in enumerate(lens):
+        s0, s1 = starts[i], starts[i + 1]
+        ref = sdpa_prefill_ref(
+            q[s0:s1].transpose(0, 1).unsqueeze(0),
+            k[s0:s1].transpose(0, 1).unsqueeze(0),
+            v[s0:s1].transpose(0, 1).unsqueeze(0), scale, causal=True)
+        ref = ref[0].transpose(0, 1).reshape(L, Hq * D)
+        assert_close_bf16(out[s0:s1], ref, atol=3e-2, rtol=3e-2,
+                          msg=f"seq {i}")
+
+
+@gpu
+@requires_gpu
+def test_prefill_attention_strided_views():
+    """q/k/v as strided views into a fused qkv buffer (the engine path)."""
+    import lws_amd.ops as ops
+    from lws_amd.ops.reference import sdpa_prefill_ref
+
+    Hq, Hkv, D, T = 8, 2, 128, 96
+    W = (Hq + 2 * Hkv) * D
+    qkv = torch.randn(T, W, dtype=torch.bfloat16, device="cuda")
+    q = qkv.narrow(-1, 0, Hq * D).unflatten(-1, (Hq, D))
+    k = qkv.narrow(-1, Hq * D, Hkv * D).unflatten(-1, (Hkv, D))
+    v = qkv.narrow(-1, (Hq + Hkv) * D, Hkv * D).unflatten(-1, (Hkv, D))
+    scale = 1.0 / math.sqrt(D)
+    out = ops.prefill_attention(q, k, v, [0, T], scale)
+    ref = sdpa_prefill_ref(q.transpose(0, 1).unsqueeze(0).contiguous(),
+                           k.transpose(0, 1).unsqueeze(0).contiguous(),
+                           v.transpose(0, 1).unsqueeze(0).contiguous(),
+                           scale, causal=True)
+    ref = ref[0].transpose(0, 1).reshape(T, Hq * D)
+    assert_close_bf16(out, ref, atol=3e-2, rtol=3e-2)
