@@ -46,8 +46,12 @@ class Sequence:
 
 
 class BlockAllocator:
+    """Page 0 is reserved as a scratch page: hipGraph-captured decode pads
+    batches to a bucket size, and padded slots write their (ignored) KV
+    there."""
+
     def __init__(self, num_pages: int):
-        self.free = list(range(num_pages - 1, -1, -1))
+        self.free = list(range(num_pages - 1, 0, -1))
 
     def alloc(self, n: int) -> list[int]:
         if n > len(self.free):
@@ -73,6 +77,9 @@ class Engine:
         self.sequences: dict[int, Sequence] = {}
         self._next_seq_id = 0
         self.ready = False
+        # hipGraph-captured decode steps keyed by (batch_bucket, page_bucket)
+        self._graphs: dict[tuple[int, int], "_CapturedDecode"] = {}
+        self.use_graphs = self.device.type == "cuda"
 
     # -- lifecycle ------------------------------------------------------
     def load(self) -> dict:
@@ -109,6 +116,7 @@ class Engine:
         self.finish(sid)
 
     def unload(self) -> None:
+        self._graphs.clear()
         self.kv_caches = []
         for s in list(self.sequences.values()):
             self.allocator.release(s.pages)
@@ -198,16 +206,22 @@ class Engine:
             slots.append(s.pages[t // PAGE_SIZE] * PAGE_SIZE + t % PAGE_SIZE)
             lens.append(t + 1)
             bt[i, :len(s.pages)] = torch.tensor(s.pages, dtype=torch.int32)
-        batch = DecodeBatch(
-            input_ids=torch.tensor(ids, device=self.device, dtype=torch.long),
-            positions=torch.tensor(pos, device=self.device, dtype=torch.int32),
-            block_tables=bt.to(self.device),
-            seq_lens=torch.tensor(lens, device=self.device, dtype=torch.int32),
-            slot_mapping=torch.tensor(slots, device=self.device,
-                                      dtype=torch.int64))
-        hidden = self.model.forward_decode(batch, self.kv_caches)
-        logits = self.model.compute_logits(hidden)
-        next_tokens = logits.argmax(dim=-1)
+        if self.use_graphs:
+            next_tokens = self._decode_graphed(ids, pos, bt, lens, slots)
+        else:
+            batch = DecodeBatch(
+                input_ids=torch.tensor(ids, device=self.device,
+                                       dtype=torch.long),
+                positions=torch.tensor(pos, device=self.device,
+                                       dtype=torch.int32),
+                block_tables=bt.to(self.device),
+                seq_lens=torch.tensor(lens, device=self.device,
+                                      dtype=torch.int32),
+                slot_mapping=torch.tensor(slots, device=self.device,
+                                          dtype=torch.int64))
+            hidden = self.model.forward_decode(batch, self.kv_caches)
+            logits = self.model.compute_logits(hidden)
+            next_tokens = logits.argmax(dim=-1)
         out = {}
         for i, s in enumerate(seqs):
             tok = int(next_tokens[i])
@@ -215,6 +229,23 @@ class Engine:
             s.token_ids.append(tok)
             out[s.seq_id] = tok
         return out
+
+    def _decode_graphed(self, ids, pos, bt, lens, slots) -> torch.Tensor:
+        """Replay (or capture) the hipGraph for this batch/page bucket."""
+        B = len(ids)
+        b_bucket = 1
+        while b_bucket < B:
+            b_bucket *= 2
+        b_bucket = min(b_bucket, self.cfg.max_batch)
+        p_bucket = 8
+        while p_bucket < bt.size(1):
+            p_bucket *= 2
+        key = (b_bucket, p_bucket)
+        g = self._graphs.get(key)
+        if g is None:
+            g = _CapturedDecode(self, b_bucket, p_bucket)
+            self._graphs[key] = g
+        return g.run(ids, pos, bt, lens, slots)[:B]
 
     # -- convenience ----------------------------------------------------
     def generate(self, prompts: list[list[int]],
@@ -229,3 +260,75 @@ class Engine:
             outs.append(seq.token_ids[plens[sid]:plens[sid] + max_new_tokens])
             self.finish(sid)
         return outs
+
+
+class _CapturedDecode:
+    """One hipGraph-captured decode step for a (batch, pages) bucket.
+
+    Static input tensors are refreshed on the host side before each
+    replay; padded batch slots decode token 0 at position 0 against the
+    reserved scratch page (their outputs are dropped).  Falls back to
+    eager execution if capture fails (e.g. an uncapturable collective),
+    running the same kernels either way.
+    """
+
+    def __init__(self, engine: Engine, B: int, max_pages: int):
+        self.engine = engine
+        self.B = B
+        dev = engine.device
+        self.ids = torch.zeros(B, dtype=torch.long, device=dev)
+        self.pos = torch.zeros(B, dtype=torch.int32, device=dev)
+        self.bt = torch.zeros(B, max_pages, dtype=torch.int32, device=dev)
+        self.lens = torch.ones(B, dtype=torch.int32, device=dev)
+        self.slots = torch.zeros(B, dtype=torch.int64, device=dev)
+        self.tokens = torch.zeros(B, dtype=torch.long, device=dev)
+        self.batch = DecodeBatch(input_ids=self.ids, positions=self.pos,
+                                 block_tables=self.bt, seq_lens=self.lens,
+                                 slot_mapping=self.slots)
+        self.graph = None
+        self._capture()
+
+    def _forward(self) -> None:
+        eng = self.engine
+        hidden = eng.model.forward_decode(self.batch, eng.kv_caches)
+        logits = eng.model.compute_logits(hidden)
+        torch.argmax(logits, dim=-1, out=self.tokens)
+
+    def _capture(self) -> None:
+        import logging
+
+        eng = self.engine
+        try:
+            # warm up twice outside capture (allocator + lazy inits)
+            for _ in range(2):
+                self._forward()
+            torch.cuda.synchronize(eng.device)
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self._forward()
+        except Exception as e:  # noqa: BLE001 — capture-unsupported path
+            logging.getLogger("lws_amd").warning(
+                "hipGraph capture failed (%s); decode runs eager", e)
+            self.graph = None
+
+    def run(self, ids, pos, bt, lens, slots) -> torch.Tensor:
+        B = len(ids)
+        dev = self.engine.device
+        self.ids[:B].copy_(torch.tensor(ids, dtype=torch.long), non_blocking=True)
+        if B < self.B:
+            self.ids[B:].zero_()
+            self.pos[B:].zero_()
+            self.lens[B:].fill_(1)
+            self.bt[B:].zero_()
+            self.slots[B:] = torch.arange(self.B - B, device=dev) % 16
+        self.pos[:B].copy_(torch.tensor(pos, dtype=torch.int32), non_blocking=True)
+        self.bt[:B, :bt.size(1)].copy_(bt, non_blocking=True)
+        if bt.size(1) < self.bt.size(1):
+            self.bt[:B, bt.size(1):].zero_()
+        self.lens[:B].copy_(torch.tensor(lens, dtype=torch.int32), non_blocking=True)
+        self.slots[:B].copy_(torch.tensor(slots, dtype=torch.int64), non_blocking=True)
+        if self.graph is not None:
+            self.graph.replay()
+        else:
+            self._forward()
+        return self.tokens

@@ -126,3 +126,21 @@ def test_mixtral_engine_cpu():
     eng2.load()
     out2 = eng2.generate([prompt + out[:2]], max_new_tokens=1)[0]
     assert out2[0] == out[2]
+
+
+@gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+def test_engine_gpu_graph_decode_matches_eager():
+    """hipGraph-captured decode must produce the same tokens as eager."""
+    eng = _make_engine(device="cuda", seed=11)
+    assert eng.use_graphs
+    prompt = [5, 17, 250, 3, 99]
+    out_g = eng.generate([list(prompt), [7, 8, 9]], max_new_tokens=6)
+    assert eng._graphs, "no graph captured"
+    assert any(g.graph is not None for g in eng._graphs.values()), \
+        "capture fell back to eager"
+
+    eng2 = _make_engine(device="cuda", seed=11)
+    eng2.use_graphs = False
+    out_e = eng2.generate([list(prompt), [7, 8, 9]], max_new_tokens=6)
+    assert out_g == out_e
