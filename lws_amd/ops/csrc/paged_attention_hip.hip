@@ -24,6 +24,10 @@
 #define PA_NWAVES 4
 #define PA_MAX_GQA 16
 
+// G is a template parameter so the per-head loops fully unroll and the
+// accumulator arrays stay in VGPRs — runtime-indexed register arrays are
+// demoted to scratch memory (guide rule #20), which costs ~25x here.
+template <int G>
 __global__ __launch_bounds__(256)
 void paged_attention_chunk_kernel(
     float* __restrict__ ws_acc,      // [B, Hkv, chunks, G, 128]
@@ -33,7 +37,7 @@ void paged_attention_chunk_kernel(
     const ushort* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [B, max_pages]
     const int* __restrict__ seq_lens,      // [B]
-    float scale, int G, int Hkv, int page_size, int max_pages,
+    float scale, int Hkv, int page_size, int max_pages,
     int chunk_keys, int num_chunks, long long q_stride) {
   const int b = blockIdx.x;
   const int hkv = blockIdx.y;
@@ -61,6 +65,7 @@ void paged_attention_chunk_kernel(
 
   // q for the whole GQA group staged in LDS: [G][128] fp32 (pre-scaled)
   __shared__ float q_lds[PA_MAX_GQA][PA_HEAD_DIM];
+#pragma unroll
   for (int g = 0; g < G; ++g) {
     const ushort* qrow = q + (long long)b * q_stride
         + ((long long)hkv * G + g) * PA_HEAD_DIM;
@@ -73,9 +78,16 @@ void paged_attention_chunk_kernel(
   }
   __syncthreads();
 
-  // per-wave online-softmax state + per-thread V accumulator
-  float m_run[PA_MAX_GQA], l_run[PA_MAX_GQA];
-  float acc[PA_MAX_GQA][8];
+  // per-wave online-softmax state + per-thread V accumulator.
+  // Defer-max (guide T13): the running max only moves when a score
+  // exceeds it by THR, so the common pass skips the O(G*8) rescale and
+  // the cross-group max shuffles; exp values are bounded by e^THR.
+  // l accumulates per lane-group and is reduced across groups once at
+  // the end (contributions share m_run between rescale events).
+  const float PA_THR = 8.0f;
+  float m_run[G], l_run[G];
+  float acc[G][8];
+#pragma unroll
   for (int g = 0; g < G; ++g) {
     m_run[g] = -INFINITY;
     l_run[g] = 0.0f;
@@ -105,43 +117,55 @@ void paged_attention_chunk_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) kf[j] = 0.0f;
     }
-    // scores for all G heads (dot over this lane's slice, reduced across
-    // the 16 lanes of the key group), then per-wave online softmax
+    // V row loaded once per pass, reused by all G heads
     float vf[8];
-    bool v_loaded = false;
+    if (valid) {
+      bf16x8 vv;
+      vv.u = *reinterpret_cast<const uint4*>(v_cache + row + slice * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.h[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vf[j] = 0.0f;
+    }
+    // scores for all G heads (dot over this lane's slice, reduced across
+    // the 16 lanes of the key group), then deferred online softmax
+#pragma unroll
     for (int g = 0; g < G; ++g) {
       float p = 0.0f;
 #pragma unroll
       for (int j = 0; j < 8; ++j) p += kf[j] * q_lds[g][slice * 8 + j];
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1) p += __shfl_xor(p, off, WAVE_SIZE);
-      float s_local = valid ? p : -INFINITY;
-      // wave-wide max over the 4 key groups
-      float m4 = s_local;
-      m4 = fmaxf(m4, __shfl_xor(m4, 16, WAVE_SIZE));
-      m4 = fmaxf(m4, __shfl_xor(m4, 32, WAVE_SIZE));
-      const float m_new = fmaxf(m_run[g], m4);
-      if (m_new == -INFINITY) continue;  // nothing valid yet for this wave
-      const float rescale =
-          (m_run[g] == -INFINITY) ? 0.0f : __expf(m_run[g] - m_new);
-      const float e = (s_local == -INFINITY) ? 0.0f : __expf(s_local - m_new);
-      float l4 = e;
-      l4 += __shfl_xor(l4, 16, WAVE_SIZE);
-      l4 += __shfl_xor(l4, 32, WAVE_SIZE);
-      l_run[g] = l_run[g] * rescale + l4;
-      m_run[g] = m_new;
-      if (!v_loaded && valid) {
-        bf16x8 vv;
-        vv.u = *reinterpret_cast<const uint4*>(v_cache + row + slice * 8);
+      const float s_local = valid ? p : -INFINITY;
+      // slow path (rare): a score moved past the deferred max budget
+      if (__any(s_local > m_run[g] + PA_THR)) {
+        float m4 = s_local;
+        m4 = fmaxf(m4, __shfl_xor(m4, 16, WAVE_SIZE));
+        m4 = fmaxf(m4, __shfl_xor(m4, 32, WAVE_SIZE));
+        const float rescale =
+            (m_run[g] == -INFINITY) ? 0.0f : __expf(m_run[g] - m4);
+        l_run[g] *= rescale;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) vf[j] = bf16_to_f32(vv.h[j]);
-        v_loaded = true;
+        for (int j = 0; j < 8; ++j) acc[g][j] *= rescale;
+        m_run[g] = m4;
       }
+      const float e = (s_local == -INFINITY) ? 0.0f
+                                             : __expf(s_local - m_run[g]);
+      l_run[g] += e;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        acc[g][j] = acc[g][j] * rescale + (valid ? e * vf[j] : 0.0f);
-      }
+      for (int j = 0; j < 8; ++j) acc[g][j] += e * vf[j];
     }
+  }
+
+  // l_run holds per-lane-group partial sums; total them across the wave's
+  // 4 key groups (each group's 16 lanes agree on its value)
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    float l4 = l_run[g];
+    l4 += __shfl_xor(l4, 16, WAVE_SIZE);
+    l4 += __shfl_xor(l4, 32, WAVE_SIZE);
+    l_run[g] = l4;
   }
 
   // merge the 4 waves' (m, l, acc) through LDS with softmax rescaling.
@@ -293,14 +317,25 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 
   hipStream_t stream = at::hip::getCurrentHIPStream();
   dim3 grid(B, Hkv, num_chunks);
-  hipLaunchKernelGGL(paged_attention_chunk_kernel, grid, dim3(256), 0, stream,
-                     ws_acc.data_ptr<float>(), ws_ml.data_ptr<float>(),
-                     (const ushort*)q.data_ptr(),
-                     (const ushort*)k_cache.data_ptr(),
-                     (const ushort*)v_cache.data_ptr(),
-                     block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                     (float)scale, G, Hkv, page_size, max_pages,
-                     (int)chunk_keys, num_chunks, (long long)q.stride(0));
+#define PA_LAUNCH(GG)                                                        \
+  hipLaunchKernelGGL((paged_attention_chunk_kernel<GG>), grid, dim3(256), 0,  \
+                     stream, ws_acc.data_ptr<float>(), ws_ml.data_ptr<float>(),\
+                     (const ushort*)q.data_ptr(),                             \
+                     (const ushort*)k_cache.data_ptr(),                       \
+                     (const ushort*)v_cache.data_ptr(),                       \
+                     block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),  \
+                     (float)scale, Hkv, page_size, max_pages,                 \
+                     (int)chunk_keys, num_chunks, (long long)q.stride(0))
+  switch (G) {
+    case 1: PA_LAUNCH(1); break;
+    case 2: PA_LAUNCH(2); break;
+    case 4: PA_LAUNCH(4); break;
+    case 8: PA_LAUNCH(8); break;
+    case 16: PA_LAUNCH(16); break;
+    default:
+      TORCH_CHECK(false, "paged_attention: GQA group must be 1/2/4/8/16");
+  }
+#undef PA_LAUNCH
   hipLaunchKernelGGL(paged_attention_reduce_kernel, dim3(B, Hq), dim3(128), 0,
                      stream, (ushort*)out.data_ptr(), ws_acc.data_ptr<float>(),
                      ws_ml.data_ptr<float>(), seq_lens.data_ptr<int>(),
