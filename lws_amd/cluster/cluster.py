@@ -67,10 +67,13 @@ class LwsCluster:
                 self.agents.append(NodeAgent(self.manager, node, rt))
 
         # LWS controllers
-        self.lws_reconciler = LeaderWorkerSetReconciler(self.manager)
+        from .events import EventRecorder
+        self.recorder = EventRecorder(self.store)
+        self.lws_reconciler = LeaderWorkerSetReconciler(self.manager,
+                                                        recorder=self.recorder)
         self.pod_reconciler = PodReconciler(
             self.manager, scheduler_provider=self.scheduler_provider,
-            node_lookup=self._node_by_name.get)
+            node_lookup=self._node_by_name.get, recorder=self.recorder)
 
         # DisaggregatedSet controller suite
         self.ds_reconciler = None
@@ -78,7 +81,8 @@ class LwsCluster:
             try:
                 from ..controllers.disaggregatedset.controller import (
                     DisaggregatedSetReconciler)
-                self.ds_reconciler = DisaggregatedSetReconciler(self.manager)
+                self.ds_reconciler = DisaggregatedSetReconciler(
+                    self.manager, recorder=self.recorder)
             except ImportError:
                 pass  # DS suite not built yet (round-1 staging)
 

@@ -151,6 +151,7 @@ class Manager:
     def __init__(self, store: Optional[Store] = None) -> None:
         self.store = store or Store()
         self.controllers: list[Controller] = []
+        self._watches: list[tuple[str, Callable[[str, object], None]]] = []
 
     def add_controller(self, ctrl: Controller) -> Controller:
         self.controllers.append(ctrl)
@@ -166,10 +167,18 @@ class Manager:
                 for ns, name in map_fn(event, obj) or []:
                     ctrl.enqueue(ns, name)
         self.store.add_handler(kind, handler)
+        self._watches.append((kind, handler))
 
     def start(self) -> None:
         for c in self.controllers:
             c.start()
+        # initial informer sync: replay ADDED for every existing object so a
+        # manager started over a pre-populated store (controller restart)
+        # reconciles everything — this is what lets a restarted manager
+        # resume a rollout mid-flight (controller-runtime cache sync)
+        for kind, handler in self._watches:
+            for obj in self.store.list(kind):
+                handler("ADDED", obj)
 
     def stop(self) -> None:
         for c in self.controllers:

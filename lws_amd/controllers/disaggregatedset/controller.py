@@ -24,12 +24,13 @@ from .service_manager import ServiceManager
 
 
 class DisaggregatedSetReconciler:
-    def __init__(self, manager: Manager) -> None:
+    def __init__(self, manager: Manager, recorder=None) -> None:
         self.store: Store = manager.store
         self.lws_manager = LeaderWorkerSetManager(self.store)
         self.service_manager = ServiceManager(self.store)
         self.scaler_manager = ScalerManager(self.store)
-        self.executor = RollingUpdateExecutor(self.lws_manager)
+        self.executor = RollingUpdateExecutor(self.lws_manager,
+                                              record=recorder)
         self.ctrl = Controller("disaggregatedset", self.reconcile)
         manager.add_controller(self.ctrl)
         manager.watch(dsapi.KIND, self.ctrl)

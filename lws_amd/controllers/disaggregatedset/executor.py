@@ -107,6 +107,9 @@ class RollingUpdateExecutor:
     def _init_rolling_update(self, ds, slice_, revision, role_names,
                              role_configs, old_revisions) -> float:
         """executor.go:87-127."""
+        if self.record is not None:
+            self.record.eventf(ds, "Normal", "RollingUpdateStarted",
+                               f"Started rolling update to revision {revision}")
         for rev in old_revisions:
             for lws in rev.roles.values():
                 replicas = dsutils.get_lws_replicas(lws)
@@ -148,6 +151,10 @@ class RollingUpdateExecutor:
         step = compute_next_step(initial_old, current_old, current_new,
                                  target_new, config)
         if step is None:
+            if self.record is not None:
+                self.record.eventf(ds, "Normal", "RollingUpdateCompleted",
+                                   "Completed rolling update to revision "
+                                   f"{new_revision.revision}")
             return None  # rollout complete
         self._scale_up_new(ds, slice_, new_revision, all_role_names, spec_set,
                            current_new, step.new)

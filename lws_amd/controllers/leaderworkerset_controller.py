@@ -146,8 +146,10 @@ def sort_by_index(items, index_fn, length: int):
 
 
 class LeaderWorkerSetReconciler:
-    def __init__(self, manager: Manager) -> None:
+    def __init__(self, manager: Manager, recorder=None) -> None:
+        from ..cluster.events import NullRecorder
         self.store: Store = manager.store
+        self.record = recorder or NullRecorder()
         self.ctrl = Controller("leaderworkerset", self.reconcile)
         self._revision_cache: dict = {}
         manager.add_controller(self.ctrl)
@@ -183,12 +185,28 @@ class LeaderWorkerSetReconciler:
         lws_updated = updated_revision is not None
         if lws_updated:
             revision = revisionutils.create_revision(self.store, updated_revision)
+            self.record.eventf(lws, "Normal", "CreatingRevision",
+                               f"Creating revision with key "
+                               f"{revisionutils.get_revision_key(revision)} "
+                               "for updated LWS")
         revision_key = revisionutils.get_revision_key(revision)
 
         partition, replicas = self._rolling_update_parameters(
             lws, leader_sts, revision_key, lws_updated)
 
+        old_partition = None
+        if leader_sts is not None and \
+                leader_sts.spec.update_strategy.rolling_update is not None:
+            old_partition = leader_sts.spec.update_strategy.rolling_update.partition
         self._ssa_leader_statefulset(lws, partition, replicas, revision_key)
+        if leader_sts is None:
+            self.record.eventf(lws, "Normal", "GroupsProgressing",
+                               f"Created leader statefulset {lws.metadata.name}")
+        elif not lws_updated and old_partition is not None and \
+                partition != old_partition:
+            self.record.eventf(lws, "Normal", "GroupsUpdating",
+                               f"Updating replicas {partition} to "
+                               f"{old_partition - 1} (inclusive)")
         self._reconcile_headless_services(lws)
 
         update_done = self._update_status(lws, revision_key)

@@ -28,8 +28,10 @@ from .leaderworkerset_controller import _merge_metadata
 
 class PodReconciler:
     def __init__(self, manager: Manager, scheduler_provider=None,
-                 node_lookup=None) -> None:
+                 node_lookup=None, recorder=None) -> None:
+        from ..cluster.events import NullRecorder
         self.store: Store = manager.store
+        self.record = recorder or NullRecorder()
         self.scheduler_provider = scheduler_provider
         # node_lookup(node_name) -> Node (for topology label resolution)
         self.node_lookup = node_lookup or (lambda name: None)
@@ -155,6 +157,11 @@ class PodReconciler:
                               leader.metadata.name, propagation="Foreground")
         except NotFoundError:
             pass
+        self.record.eventf(
+            lws, "Normal", "RecreateGroup",
+            f"Worker pod {pod.metadata.name} failed, deleted leader pod "
+            f"{leader.metadata.name} to recreate group "
+            f"{(leader.metadata.labels or {}).get(lwsapi.GROUP_INDEX_LABEL_KEY, '')}")
         return True
 
     def _worker_pod_belongs_to_leader(self, pod, leader) -> bool:
