@@ -195,7 +195,42 @@ class StatefulSetController:
         try:
             self.store.create(pod)
         except AlreadyExistsError:
-            pass
+            return
+        self._ensure_pvcs(sts, pod)
+
+    def _ensure_pvcs(self, sts: StatefulSet, pod: Pod) -> None:
+        """PVCs from volumeClaimTemplates, named <claim>-<pod> (k8s STS
+        convention; reference controller_utils.go:67-96 builds the same
+        claims from lws.volumeClaimTemplates).  Retention policy: Delete
+        whenScaled -> owned by the pod; Delete whenDeleted -> owned by the
+        StatefulSet; Retain (default) -> standalone, survives both."""
+        from ..api import serde as _serde
+
+        created = self.store.try_get("Pod", pod.metadata.namespace,
+                                     pod.metadata.name)
+        if created is None:
+            return
+        policy = sts.spec.persistent_volume_claim_retention_policy
+        for vct in sts.spec.volume_claim_templates or []:
+            pvc = _serde.deep_copy(vct)
+            pvc.metadata.name = f"{vct.metadata.name}-{pod.metadata.name}"
+            pvc.metadata.namespace = sts.metadata.namespace
+            pvc.metadata.labels = dict(vct.metadata.labels or {})
+            pvc.metadata.labels[STS_OWNER_LABEL] = sts.metadata.name
+            pvc.metadata.owner_references = []
+            if policy is not None and policy.when_scaled == "Delete":
+                pvc.metadata.owner_references = [OwnerReference(
+                    api_version="v1", kind="Pod", name=created.metadata.name,
+                    uid=created.metadata.uid, controller=False)]
+            elif policy is not None and policy.when_deleted == "Delete":
+                pvc.metadata.owner_references = [OwnerReference(
+                    api_version=sts.api_version, kind=sts.kind,
+                    name=sts.metadata.name, uid=sts.metadata.uid,
+                    controller=False)]
+            try:
+                self.store.create(pvc)
+            except AlreadyExistsError:
+                pass
 
     def _delete_pod(self, pod: Pod) -> None:
         if pod.metadata.deletion_timestamp is not None:

@@ -151,3 +151,30 @@ def test_scale_up_and_down(cluster):
              desc="scale down to 2 pods", timeout=30)
     names = sorted(p.metadata.name for p in cluster.store.list("Pod", "default"))
     assert names == ["scale-0", "scale-0-1"]
+
+
+def test_volume_claim_templates_create_pvcs(cluster):
+    from lws_amd.api.core import (PersistentVolumeClaim,
+                                  PersistentVolumeClaimSpec,
+                                  ResourceRequirements,
+                                  StatefulSetPersistentVolumeClaimRetentionPolicy)
+    from lws_amd.api.meta import ObjectMeta
+
+    lws = make_lws(name="pvc", replicas=1, size=2)
+    vct = PersistentVolumeClaim(
+        metadata=ObjectMeta(name="model-cache"),
+        spec=PersistentVolumeClaimSpec(
+            access_modes=["ReadWriteOnce"],
+            resources=ResourceRequirements(requests={"storage": "10Gi"})))
+    lws.spec.leader_worker_template.volume_claim_templates = [vct]
+    lws.spec.leader_worker_template.persistent_volume_claim_retention_policy = \
+        StatefulSetPersistentVolumeClaimRetentionPolicy(when_deleted="Delete")
+    cluster.store.create(lws)
+
+    def pvcs():
+        got = cluster.store.list("PersistentVolumeClaim", "default")
+        return got if len(got) == 2 else None
+    got = wait_for(pvcs, desc="2 PVCs (leader + worker)", timeout=20)
+    names = sorted(p.metadata.name for p in got)
+    assert names == ["model-cache-pvc-0", "model-cache-pvc-0-1"]
+    assert got[0].spec.resources.requests["storage"] == "10Gi"
