@@ -70,7 +70,7 @@ __device__ __forceinline__ void sg_stage_async(
   }
 }
 
-template <int MTILES>
+template <int MTILES, bool XLDS>
 __global__ __launch_bounds__(256)
 void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
                         const ushort* __restrict__ x,    // [M, K]
@@ -86,7 +86,7 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
   const int frag_kgrp = lane / 16;       // which 8-wide k group
 
   __shared__ ushort w_lds[2][SG_ROWS * SG_KSUB];
-  __shared__ ushort x_lds[2][16 * MTILES * SG_KSUB];
+  __shared__ ushort x_lds[XLDS ? 2 : 1][XLDS ? 16 * MTILES * SG_KSUB : 1];
 
   f32x4_t acc[MTILES];
 #pragma unroll
@@ -99,7 +99,8 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
   // prologue: stage sub-slice 0 into buffer 0
   sg_stage_async(w_lds[0], w_base + kbegin, K, SG_ROWS, w_rows_valid, wave,
                  lane);
-  sg_stage_async(x_lds[0], x + kbegin, K, 16 * MTILES, M, wave, lane);
+  if (XLDS)
+    sg_stage_async(x_lds[0], x + kbegin, K, 16 * MTILES, M, wave, lane);
   __syncthreads();
 
   for (int s = 0; s < nsub; ++s) {
@@ -108,12 +109,13 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
       const int ks_next = kbegin + (s + 1) * SG_KSUB;
       sg_stage_async(w_lds[cur ^ 1], w_base + ks_next, K, SG_ROWS,
                      w_rows_valid, wave, lane);
-      sg_stage_async(x_lds[cur ^ 1], x + ks_next, K, 16 * MTILES, M, wave,
-                     lane);
+      if (XLDS)
+        sg_stage_async(x_lds[cur ^ 1], x + ks_next, K, 16 * MTILES, M, wave,
+                       lane);
     }
     // compute current sub-slice: 4 k-steps of 32
     const ushort* wt = w_lds[cur];
-    const ushort* xt = x_lds[cur];
+    const ushort* xt = XLDS ? x_lds[cur & (XLDS ? 1 : 0)] : nullptr;
 #pragma unroll
     for (int k0 = 0; k0 < SG_KSUB; k0 += 32) {
       const int colb = (k0 + frag_kgrp * 8) * 2;
@@ -133,9 +135,18 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
         bf16x8_t afrag;
         {
           bf16x8 tmp;
-          tmp.u = *reinterpret_cast<const uint4*>(
-              reinterpret_cast<const char*>(xt) + m * SG_ROWB +
-              sg_swz(m, colb));
+          if (XLDS) {
+            tmp.u = *reinterpret_cast<const uint4*>(
+                reinterpret_cast<const char*>(xt) + m * SG_ROWB +
+                sg_swz(m, colb));
+          } else {
+            // x is tiny and L2-resident; read the fragment from global
+            const int mm = m < M ? m : M - 1;
+            tmp.u = *reinterpret_cast<const uint4*>(
+                x + (long long)mm * K + kbegin + s * SG_KSUB + k0
+                + frag_kgrp * 8);
+            if (m >= M) tmp.u = make_uint4(0, 0, 0, 0);
+          }
 #pragma unroll
           for (int j = 0; j < 8; ++j) afrag[j] = (short)tmp.h[j];
         }
@@ -212,15 +223,18 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
               "skinny_gemm workspace too small");
 
   dim3 grid(n_blocks, grid_y);
+  const bool xlds = env_int("LWS_SG_XLDS", 1) != 0;
+#define SG_LAUNCH(MT, XL)                                                    \
+  hipLaunchKernelGGL((skinny_gemm_kernel<MT, XL>), grid, dim3(256), 0,        \
+                     stream, ws.data_ptr<float>(),                            \
+                     (const ushort*)x.data_ptr(),                             \
+                     (const ushort*)w.data_ptr(), M, N, K, k_slice)
   if (M <= 16) {
-    hipLaunchKernelGGL((skinny_gemm_kernel<1>), grid, dim3(256), 0, stream,
-                       ws.data_ptr<float>(), (const ushort*)x.data_ptr(),
-                       (const ushort*)w.data_ptr(), M, N, K, k_slice);
+    if (xlds) SG_LAUNCH(1, true); else SG_LAUNCH(1, false);
   } else {
-    hipLaunchKernelGGL((skinny_gemm_kernel<2>), grid, dim3(256), 0, stream,
-                       ws.data_ptr<float>(), (const ushort*)x.data_ptr(),
-                       (const ushort*)w.data_ptr(), M, N, K, k_slice);
+    if (xlds) SG_LAUNCH(2, true); else SG_LAUNCH(2, false);
   }
+#undef SG_LAUNCH
   long long total = (long long)M * N;
   long long blocks = min((total + 255) / 256, (long long)2048);
   hipLaunchKernelGGL(skinny_gemm_finalize_kernel, dim3((int)blocks), dim3(256),
