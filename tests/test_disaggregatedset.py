@@ -225,3 +225,76 @@ def test_ds_slices():
         wait_for(one_slice, desc="slice-1 cleanup", timeout=60)
     finally:
         c.stop()
+
+
+def test_ds_role_add_and_remove(cluster):
+    from tests.conftest import retry_update
+
+    ds = make_ds(roles=[("prefill", 1, 1), ("decode", 1, 1)])
+    cluster.store.create(ds)
+    wait_for(lambda: ds_available(cluster), desc="Available", timeout=60)
+
+    # add a third role (template change -> new revision, lockstep rollout)
+    def add_role(o):
+        import copy
+        from lws_amd.api import serde
+        new_role = serde.from_dict(type(o.spec.roles[0]),
+                                   serde.to_dict(o.spec.roles[0]))
+        new_role.name = "cache"
+        o.spec.roles.append(new_role)
+    retry_update(cluster.store, dsapi.KIND, "default", "my-ds", add_role)
+
+    def three_roles():
+        c = ds_available(cluster)
+        if c is None:
+            return None
+        names = {rs.name for rs in c.status.role_statuses}
+        return c if names == {"prefill", "decode", "cache"} else None
+    wait_for(three_roles, desc="role added + Available", timeout=120)
+
+    # remove it again; its LWS must drain away and status drop the entry
+    retry_update(cluster.store, dsapi.KIND, "default", "my-ds",
+                 lambda o: o.spec.roles.pop())
+
+    def two_roles():
+        c = ds_available(cluster)
+        if c is None:
+            return None
+        names = {rs.name for rs in c.status.role_statuses}
+        if names != {"prefill", "decode"}:
+            return None
+        lws_roles = {l.metadata.labels[dsapi.ROLE_LABEL_KEY]
+                     for l in cluster.store.list(lwsapi.KIND, "default")}
+        return c if "cache" not in lws_roles else None
+    wait_for(two_roles, desc="role removed + drained", timeout=120)
+
+
+def test_ds_exclusive_slice_placement():
+    from lws_amd.api.disaggregatedset import PlacementPolicy
+    from lws_amd.cluster.cluster import LwsCluster, make_nodes
+
+    topo = "topology.lws.amd.com/island"
+    # 2 islands x 4 GPUs; 2 slices of (prefill 1 + decode 1 pods, 1 GPU each)
+    c = LwsCluster(nodes=make_nodes(2, gpus_per_node=4)).start()
+    try:
+        ds = make_ds(roles=[("prefill", 1, 1), ("decode", 1, 1)], slices=2,
+                     placement=PlacementPolicy(type="ExclusiveSlice",
+                                               topology=topo))
+        for role in ds.spec.roles:
+            role.spec.leader_worker_template.worker_template.spec \
+                .containers[0].resources.requests = {"amd.com/gpu": 1}
+        c.store.create(ds)
+        wait_for(lambda: ds_available(c), desc="2-slice placed Available",
+                 timeout=90)
+        pods = c.store.list("Pod", "default")
+        assert len(pods) == 4
+        by_slice = {}
+        for p in pods:
+            sl = p.metadata.labels[dsapi.SLICE_LABEL_KEY]
+            node = c.node(p.node_name)
+            by_slice.setdefault(sl, set()).add(node.metadata.labels[topo])
+        # each slice's roles co-located on ONE island; slices on DIFFERENT
+        assert all(len(v) == 1 for v in by_slice.values()), by_slice
+        assert len({next(iter(v)) for v in by_slice.values()}) == 2, by_slice
+    finally:
+        c.stop()
