@@ -263,12 +263,6 @@ class LlamaForCausalLM:
             x, residual, w, self.cfg.rms_eps)
         return out, new_res
 
-    def _rope(self, q, k, positions, hq, hkv):
-        if self._ops.is_gpu:
-            self._ops.ops.rope(q, k, self.rope_table, positions, hq, hkv)
-            return q, k
-        return self._ops.ref.rope_ref(q, k, self.rope_table, positions, hq, hkv)
-
     def _silu_mul(self, gu):
         if self._ops.is_gpu:
             return self._ops.ops.silu_mul(gu)

@@ -78,8 +78,10 @@ class Engine:
         self._next_seq_id = 0
         self.ready = False
         # hipGraph-captured decode steps keyed by (batch_bucket, page_bucket)
+        import os
         self._graphs: dict[tuple[int, int], "_CapturedDecode"] = {}
-        self.use_graphs = self.device.type == "cuda"
+        self.use_graphs = (self.device.type == "cuda"
+                           and os.environ.get("LWS_AMD_NO_GRAPHS", "0") != "1")
 
     # -- lifecycle ------------------------------------------------------
     def load(self) -> dict:
