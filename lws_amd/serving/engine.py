@@ -113,8 +113,16 @@ class Engine:
 
     def _warmup(self) -> None:
         sid = self.add_request([1, 2, 3, 4])
+        t0 = time.perf_counter()
         self.step()          # prefill
-        self.step()          # one decode
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        self.step()          # one decode (captures the first hipGraph)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        t2 = time.perf_counter()
+        self.warmup_detail = {"prefill_s": t1 - t0, "decode_capture_s": t2 - t1}
         self.finish(sid)
 
     def unload(self) -> None:
