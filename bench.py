@@ -141,6 +141,8 @@ class BenchConductor:
             rev = self.runtime.drain_pending_build(timeout=0.002)
             if rev is not None:
                 spec = self.runtime.spec_from_pods(rev)
+                if spec is None:
+                    continue  # stale event for a torn-down revision
                 self.conductor.command({"op": "build", "spec": spec})
                 self.runtime.mark_revision_ready(rev)
         raise TimeoutError(f"bench: timed out waiting for {desc}")
@@ -201,8 +203,24 @@ class BenchConductor:
         # teardown
         store.delete(lwsapi.KIND, "default", "bench-lws",
                      propagation="Background")
-        wait_until(lambda: not store.list("Pod", "default"), 600,
-                   "pods drained")
+        try:
+            # keep pumping stray build events so a late-arriving revision
+            # completion can't wedge the runtime queue
+            self._serve_builds_until(
+                lambda: not store.list("Pod", "default"), 600,
+                "pods drained")
+        except TimeoutError:
+            for pod in store.list("Pod", "default"):
+                print(f"STUCK POD {pod.metadata.name} phase={pod.status.phase}"
+                      f" node={pod.node_name!r}"
+                      f" deleting={pod.metadata.deletion_timestamp is not None}"
+                      f" finalizers={pod.metadata.finalizers}",
+                      file=sys.stderr, flush=True)
+            print(f"runtime.started={ {r: list(p) for r, p in self.runtime.started.items()} }",
+                  file=sys.stderr, flush=True)
+            print(f"store keys={self.cluster.store.snapshot_keys()}",
+                  file=sys.stderr, flush=True)
+            raise
         self.conductor.command({"op": "teardown"})
         return {"time_to_ready_s": t_ready, "rollout_s": t_rollout,
                 "shard": shard_info}

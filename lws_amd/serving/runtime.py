@@ -185,9 +185,11 @@ class CollectiveGroupRuntime(PodRuntime):
             pods.pop(pod.metadata.name, None)
         agent.finish_pod_teardown(pod)
 
-    def spec_from_pods(self, rev: str) -> dict:
+    def spec_from_pods(self, rev: str) -> Optional[dict]:
         with self._lock:
             pods = self.started.get(rev, {})
+            if not pods:
+                return None  # revision torn down before its build was served
             pod, _ = next(iter(pods.values()))
         ann = pod.metadata.annotations or {}
         return {
