@@ -199,6 +199,12 @@ class BenchConductor:
         self._serve_builds_until(rolled, timeout=3600,
                                  desc="rolling update complete")
         t_rollout = time.perf_counter() - t1
+        if t_rollout < 0.05:
+            live = store.list("Pod", "default")
+            print(f"WARN: implausibly fast rollout {t_rollout * 1e3:.1f} ms; "
+                  f"old_uids={sorted(old_rev_pods)} "
+                  f"live={[(p.metadata.name, p.metadata.uid) for p in live]}",
+                  file=sys.stderr, flush=True)
 
         # teardown
         store.delete(lwsapi.KIND, "default", "bench-lws",
