@@ -344,6 +344,8 @@ def main() -> None:
                 "warmup_s": round(shard.get("warmup_s", 0), 3),
                 "decode_tokens_per_s": (round(decode["tokens_per_s"], 1)
                                         if decode else None),
+                "decode_seconds_all": (decode.get("seconds_all")
+                                       if decode else None),
                 "orchestrator": "lws_amd in-process control plane",
             },
         }
