@@ -346,6 +346,7 @@ def main() -> None:
                                         if decode else None),
                 "decode_seconds_all": (decode.get("seconds_all")
                                        if decode else None),
+                "decode_diag": decode.get("diag") if decode else None,
                 "orchestrator": "lws_amd in-process control plane",
             },
         }

@@ -86,10 +86,30 @@ class ShardHost:
                 torch.cuda.synchronize()
             passes.append(time.perf_counter() - t0)
         dt = min(passes)
+        diag = None
+        if self.device.startswith("cuda") and dt / steps > 0.03:
+            # pathologically slow decode: split host vs device time
+            diag = {"graphs": {str(k): (g.graph is not None)
+                               for k, g in eng._graphs.items()}}
+            g = next(iter(eng._graphs.values()), None)
+            if g is not None and g.graph is not None:
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for _ in range(3):
+                    g.graph.replay()
+                torch.cuda.synchronize()
+                diag["bare_replay_s"] = (time.perf_counter() - t0) / 3
+            t0 = time.perf_counter()
+            for _ in range(3):
+                eng.step()
+            t_host = time.perf_counter() - t0
+            torch.cuda.synchronize()
+            diag["full_step_s"] = (time.perf_counter() - t0) / 3
+            diag["host_submit_s"] = t_host / 3
         for s in sids:
             eng.finish(s)
         return {"decode_steps": steps, "batch": batch, "seconds": dt,
-                "seconds_all": passes,
+                "seconds_all": passes, "diag": diag,
                 "tokens_per_s": batch * steps / dt}
 
 
