@@ -124,3 +124,30 @@ def test_informer(manager_proc):
                  desc="DELETED event", timeout=20)
     finally:
         inf.stop()
+
+
+def test_ds_over_http(manager_proc):
+    proc, base = manager_proc
+    from lws_amd.client.clientset import Clientset
+    from tests.conftest import wait_for
+    from tests.test_disaggregatedset import make_ds
+
+    cs = Clientset.for_server(base)
+    ds_client = cs.disaggregated_sets("default")
+    ds_client.create(make_ds(name="http-ds",
+                             roles=[("prefill", 1, 1), ("decode", 1, 1)]))
+
+    def available():
+        cur = ds_client.get("http-ds")
+        if cur is None:
+            return None
+        cond = next((c for c in cur.status.conditions
+                     if c.type == "Available"), None)
+        return cur if cond and cond.status == "True" else None
+    wait_for(available, desc="DS Available over HTTP", timeout=60)
+    lws_names = [o.metadata.name for o in
+                 cs.leader_worker_sets("default").list()]
+    assert len(lws_names) == 2
+    ds_client.delete("http-ds")
+    wait_for(lambda: not cs.leader_worker_sets("default").list(),
+             desc="children GC'd", timeout=30)
