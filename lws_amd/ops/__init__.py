@@ -124,6 +124,10 @@ def paged_attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
         target_chunks = max(1, 1024 // max(1, B * Hkv))
         target_chunks = min(target_chunks, (max_len + 63) // 64)
         chunk_keys = -(-max_len // target_chunks)
+        if max_len >= 1024:
+            # long contexts: per-chunk fixed costs (q staging, wave merge)
+            # dominate below 128 keys (measured A/B in profiles/)
+            chunk_keys = max(chunk_keys, 128)
         chunk_keys = ((chunk_keys + 15) // 16) * 16
     num_chunks = max(1, (max_len + chunk_keys - 1) // chunk_keys)
     if workspace is None:
