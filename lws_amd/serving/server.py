@@ -64,7 +64,16 @@ class ServingLoop:
                 if not self._pending:
                     pass
                 else:
-                    out = self.engine.step()
+                    try:
+                        out = self.engine.step()
+                    except Exception as e:  # noqa: BLE001 — fail the batch,
+                        # free its resources, keep serving later requests
+                        for p in self._pending.values():
+                            self.engine.finish(p.seq_id)
+                            if not p.future.done():
+                                p.future.set_exception(e)
+                        self._pending.clear()
+                        continue
                     self.stats["steps"] += 1
                     self.stats["tokens_generated"] += len(out)
                     finished = []

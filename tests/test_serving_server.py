@@ -62,3 +62,24 @@ def test_completions_endpoint():
         assert "lws_amd_engine_requests_total 2" in r.text
     finally:
         loop.stop()
+
+
+def test_serving_loop_fails_requests_on_engine_error():
+    """KV exhaustion (or any engine error) fails the affected futures and
+    the loop keeps serving subsequent requests."""
+    from lws_amd.serving.engine import Engine, EngineConfig
+    from lws_amd.serving.server import ServingLoop
+
+    eng = Engine(EngineConfig(model="llama-tiny", kv_pages=4, device="cpu"))
+    eng.load()
+    loop = ServingLoop(eng).start()
+    try:
+        # 3 pages usable (page 0 reserved); this prompt needs 5 pages
+        f = loop.submit(list(range(70)), max_tokens=2)
+        with pytest.raises(RuntimeError, match="KV cache exhausted"):
+            f.result(timeout=30)
+        # small request still succeeds afterwards
+        f2 = loop.submit([1, 2, 3], max_tokens=2)
+        assert len(f2.result(timeout=60)) == 2
+    finally:
+        loop.stop()
