@@ -99,6 +99,9 @@ void paged_attention_chunk_kernel(
 
   // wave-interleaved key streams: wave w takes keys [kstart+w*4+kgrp],
   // stepping 16 keys per workgroup pass — no block-level sync inside.
+  // unroll 2 so the next pass's page-table + K/V loads issue under the
+  // current pass's softmax (latency cover for the 1-wave/SIMD TP8 regime)
+#pragma unroll 2
   for (int k0 = kstart + wave * 4; k0 < kend; k0 += 4 * PA_NWAVES) {
     const int key = k0 + kgrp;
     const bool valid = key < kend;

@@ -80,7 +80,12 @@ class Engine:
         # hipGraph-captured decode steps keyed by (batch_bucket, page_bucket)
         import os
         self._graphs: dict[tuple[int, int], "_CapturedDecode"] = {}
-        self.use_graphs = (self.device.type == "cuda"
+        # hipGraph capture of RCCL collectives is untested multi-rank on
+        # this stack; default graphs to TP=1 only (LWS_AMD_GRAPHS=1 forces
+        # them on for TP>1, LWS_AMD_NO_GRAPHS=1 disables everywhere)
+        graphs_ok = (cfg.tp_world == 1
+                     or os.environ.get("LWS_AMD_GRAPHS", "0") == "1")
+        self.use_graphs = (self.device.type == "cuda" and graphs_ok
                            and os.environ.get("LWS_AMD_NO_GRAPHS", "0") != "1")
 
     # -- lifecycle ------------------------------------------------------

@@ -144,3 +144,20 @@ def test_engine_gpu_graph_decode_matches_eager():
     eng2.use_graphs = False
     out_e = eng2.generate([list(prompt), [7, 8, 9]], max_new_tokens=6)
     assert out_g == out_e
+
+
+@gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+def test_mixtral_engine_gpu():
+    from lws_amd.serving.engine import Engine, EngineConfig
+
+    eng = Engine(EngineConfig(model="mixtral-tiny", kv_pages=64,
+                              device="cuda", seed=5))
+    eng.load()
+    prompt = [3, 1, 4, 1, 5]
+    out = eng.generate([prompt], max_new_tokens=3)[0]
+    eng2 = Engine(EngineConfig(model="mixtral-tiny", kv_pages=64,
+                               device="cuda", seed=5))
+    eng2.load()
+    out2 = eng2.generate([prompt + out[:2]], max_new_tokens=1)[0]
+    assert out2[0] == out[2]
