@@ -114,21 +114,32 @@ class Engine:
         t3 = time.perf_counter()
         self.ready = True
         return {"params": n_params, "weights_s": t1 - t0, "kv_s": t2 - t1,
-                "warmup_s": t3 - t2}
+                "warmup_s": t3 - t2,
+                "warmup_detail": getattr(self, "warmup_detail", None)}
 
     def _warmup(self) -> None:
+        """Readiness warmup: one eager prefill + decode.  hipGraph capture
+        is deliberately deferred to the first real decode so it doesn't
+        inflate group time-to-ready (capture is an optimization, not a
+        readiness precondition)."""
         sid = self.add_request([1, 2, 3, 4])
-        t0 = time.perf_counter()
-        self.step()          # prefill
-        if self.device.type == "cuda":
-            torch.cuda.synchronize()
-        t1 = time.perf_counter()
-        self.step()          # one decode (captures the first hipGraph)
-        if self.device.type == "cuda":
-            torch.cuda.synchronize()
-        t2 = time.perf_counter()
-        self.warmup_detail = {"prefill_s": t1 - t0, "decode_capture_s": t2 - t1}
-        self.finish(sid)
+        graphs = self.use_graphs
+        self.use_graphs = False
+        try:
+            t0 = time.perf_counter()
+            self.step()          # prefill
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            t1 = time.perf_counter()
+            self.step()          # one eager decode
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            t2 = time.perf_counter()
+            self.warmup_detail = {"prefill_s": round(t1 - t0, 4),
+                                  "decode_s": round(t2 - t1, 4)}
+            self.finish(sid)
+        finally:
+            self.use_graphs = graphs
 
     def unload(self) -> None:
         self._graphs.clear()
