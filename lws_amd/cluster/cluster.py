@@ -58,6 +58,8 @@ class LwsCluster:
         disaggregatedset_webhook.register(self.store)
 
         # substrate controllers
+        from .gc import GarbageCollector
+        self.gc = GarbageCollector(self.manager)
         self.sts_controller = StatefulSetController(self.manager)
         self.scheduler = Scheduler(self.manager, self.nodes)
         self.agents: list[NodeAgent] = []
