@@ -70,6 +70,8 @@ class Store:
     def __init__(self) -> None:
         self._lock = threading.RLock()
         self._objects: dict[tuple[str, str, str], Any] = {}
+        # secondary index: kind -> {key -> obj} (shared object refs)
+        self._by_kind: dict[str, dict[tuple[str, str, str], Any]] = {}
         self._rv = 0
         self._uid = 0
         # kind -> list of fn(event_type, obj)
@@ -96,10 +98,13 @@ class Store:
 
     def _dispatch(self, events: list[tuple[str, Any]]) -> None:
         for ev, obj in events:
+            # one shared copy per event: watch handlers only derive queue
+            # keys from it (informer handlers must not mutate the object)
+            snapshot = serde.deep_copy(obj)
             for fn in self._handlers.get(obj_kind(obj), []):
-                fn(ev, serde.deep_copy(obj))
+                fn(ev, snapshot)
             for fn in self._all_handlers:
-                fn(ev, serde.deep_copy(obj))
+                fn(ev, snapshot)
 
     # -- core verbs --------------------------------------------------------
     def create(self, obj: Any) -> Any:
@@ -124,6 +129,7 @@ class Store:
             obj.metadata.creation_timestamp = time.time()
             obj.metadata.deletion_timestamp = None
             self._objects[key] = obj
+            self._by_kind.setdefault(key[0], {})[key] = obj
             events.append((ADDED, obj))
         self._dispatch(events)
         return serde.deep_copy(obj)
@@ -146,9 +152,7 @@ class Store:
              filter_fn: Optional[Callable[[Any], bool]] = None) -> list[Any]:
         out = []
         with self._lock:
-            for (k, ns, _), obj in self._objects.items():
-                if k != kind:
-                    continue
+            for (k, ns, _), obj in self._by_kind.get(kind, {}).items():
                 if namespace is not None and ns != namespace:
                     continue
                 if label_selector is not None:
@@ -194,6 +198,7 @@ class Store:
             if hasattr(obj, "status"):
                 obj.status = serde.deep_copy(old.status)
             self._objects[key] = obj
+            self._by_kind.setdefault(key[0], {})[key] = obj
             events.append((MODIFIED, obj))
         self._dispatch(events)
         self._maybe_finish_foreground_owners()
@@ -215,6 +220,7 @@ class Store:
             stored.status = serde.deep_copy(obj.status)
             stored.metadata.resource_version = str(self._rv)
             self._objects[key] = stored
+            self._by_kind.setdefault(key[0], {})[key] = stored
             events.append((MODIFIED, stored))
         self._dispatch(events)
         return serde.deep_copy(stored)
@@ -265,6 +271,7 @@ class Store:
                     events.append((MODIFIED, obj))
                 elif not obj.metadata.finalizers:
                     del self._objects[key]
+                    self._by_kind.get(key[0], {}).pop(key, None)
                     events.append((DELETED, obj))
                     if propagation != "Orphan":
                         for dep_key in self._dependents_locked(obj.metadata.uid):
@@ -333,6 +340,7 @@ class Store:
                         self._dependents_locked(obj.metadata.uid):
                     continue
                 del self._objects[key]
+                self._by_kind.get(key[0], {}).pop(key, None)
                 events.append((DELETED, obj))
         if events:
             self._dispatch(events)

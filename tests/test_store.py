@@ -1,0 +1,141 @@
+"""Direct store-semantics unit tests: optimistic concurrency, generation,
+finalizers, foreground cascade, apply, GC index consistency, and a scale
+smoke test for the replicas envelope."""
+import time
+
+import pytest
+
+from lws_amd.api.core import Pod, StatefulSet
+from lws_amd.api.meta import ObjectMeta, OwnerReference
+from lws_amd.cluster.store import (AlreadyExistsError, ConflictError,
+                                   NotFoundError, Store)
+
+
+def make_pod(name, ns="default", owner=None, finalizers=None):
+    p = Pod()
+    p.metadata = ObjectMeta(name=name, namespace=ns)
+    if owner is not None:
+        p.metadata.owner_references = [owner]
+    if finalizers:
+        p.metadata.finalizers = list(finalizers)
+    return p
+
+
+def test_optimistic_concurrency_and_generation():
+    s = Store()
+    sts = StatefulSet()
+    sts.metadata = ObjectMeta(name="a", namespace="default")
+    sts.spec.replicas = 1
+    created = s.create(sts)
+    assert created.metadata.generation == 1
+
+    # status update doesn't bump generation but bumps resourceVersion
+    created.status.replicas = 1
+    after_status = s.update_status(created)
+    assert after_status.metadata.generation == 1
+    assert after_status.metadata.resource_version != \
+        created.metadata.resource_version
+
+    # stale write conflicts
+    created.spec.replicas = 2
+    with pytest.raises(ConflictError):
+        s.update(created)
+
+    fresh = s.get("StatefulSet", "default", "a")
+    fresh.spec.replicas = 2
+    updated = s.update(fresh)
+    assert updated.metadata.generation == 2
+
+    with pytest.raises(AlreadyExistsError):
+        s.create(sts)
+
+
+def test_finalizer_blocks_removal():
+    s = Store()
+    s.create(make_pod("p", finalizers=["x/y"]))
+    s.delete("Pod", "default", "p")
+    assert s.try_get("Pod", "default", "p") is not None  # terminating
+    assert s.get("Pod", "default", "p").metadata.deletion_timestamp
+    s.remove_finalizer("Pod", "default", "p", "x/y")
+    assert s.try_get("Pod", "default", "p") is None
+
+
+def test_foreground_cascade_order():
+    s = Store()
+    parent = s.create(make_pod("parent"))
+    ref = OwnerReference(api_version="v1", kind="Pod", name="parent",
+                        uid=parent.metadata.uid, controller=True,
+                        block_owner_deletion=True)
+    s.create(make_pod("child", owner=ref, finalizers=["hold"]))
+    s.delete("Pod", "default", "parent", propagation="Foreground")
+    # parent waits for the child's finalizer
+    assert s.get("Pod", "default", "parent").metadata.deletion_timestamp
+    assert s.get("Pod", "default", "child").metadata.deletion_timestamp
+    s.remove_finalizer("Pod", "default", "child", "hold")
+    assert s.try_get("Pod", "default", "child") is None
+    assert s.try_get("Pod", "default", "parent") is None
+
+
+def test_orphan_propagation():
+    s = Store()
+    parent = s.create(make_pod("parent"))
+    ref = OwnerReference(api_version="v1", kind="Pod", name="parent",
+                        uid=parent.metadata.uid, controller=True)
+    s.create(make_pod("child", owner=ref))
+    s.delete("Pod", "default", "parent", propagation="Orphan")
+    assert s.try_get("Pod", "default", "parent") is None
+    assert s.try_get("Pod", "default", "child") is not None
+
+
+def test_apply_preserves_status_and_uid():
+    s = Store()
+    sts = StatefulSet()
+    sts.metadata = ObjectMeta(name="x", namespace="default")
+    sts.spec.replicas = 1
+    created = s.create(sts)
+    created.status.ready_replicas = 1
+    s.update_status(created)
+
+    newer = StatefulSet()
+    newer.metadata = ObjectMeta(name="x", namespace="default")
+    newer.spec.replicas = 5
+    applied = s.apply(newer)
+    assert applied.spec.replicas == 5
+    assert applied.metadata.uid == created.metadata.uid
+    assert applied.status.ready_replicas == 1
+
+
+def test_kind_index_consistency():
+    s = Store()
+    for i in range(20):
+        s.create(make_pod(f"p{i}"))
+    assert len(s.list("Pod")) == 20
+    for i in range(0, 20, 2):
+        s.delete("Pod", "default", f"p{i}")
+    assert len(s.list("Pod")) == 10
+    assert all(int(p.metadata.name[1:]) % 2 == 1 for p in s.list("Pod"))
+    assert s.list("StatefulSet") == []
+
+
+def test_scale_envelope_smoke():
+    """200 groups x size 2 converge in bounded time (scale envelope)."""
+    from lws_amd.cluster.cluster import LwsCluster, make_nodes
+    from tests.conftest import lws_condition, make_lws, wait_for
+
+    c = LwsCluster(nodes=make_nodes(1, gpus_per_node=8), enable_ds=False).start()
+    try:
+        t0 = time.monotonic()
+        lws = make_lws(name="big", replicas=200, size=2)
+        c.store.create(lws)
+
+        def available():
+            cur = c.get_lws("default", "big")
+            cond = lws_condition(cur, "Available")
+            return cur if cond is not None and cond.status == "True" else None
+        wait_for(available, desc="200 groups Available", timeout=120)
+        elapsed = time.monotonic() - t0
+        assert len(c.store.list("Pod", "default")) == 400
+        # record convergence time in the assertion message for visibility
+        assert elapsed < 120, elapsed
+    finally:
+        c.stop()
