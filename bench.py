@@ -341,7 +341,9 @@ def main() -> None:
                 "rollout_ms_all": [round(v, 2) for v in rollout_ms],
                 "shard_params": shard.get("params"),
                 "weights_s": round(shard.get("weights_s", 0), 3),
+                "kv_s": round(shard.get("kv_s", 0), 3),
                 "warmup_s": round(shard.get("warmup_s", 0), 3),
+                "warmup_detail": shard.get("warmup_detail"),
                 "decode_tokens_per_s": (round(decode["tokens_per_s"], 1)
                                         if decode else None),
                 "decode_seconds_all": (decode.get("seconds_all")
