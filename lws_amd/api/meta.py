@@ -8,7 +8,6 @@ uses metav1.ObjectMeta / metav1.Condition throughout.
 """
 from __future__ import annotations
 
-import dataclasses
 import time
 from dataclasses import dataclass, field
 from typing import Optional, Union

@@ -12,14 +12,14 @@ from __future__ import annotations
 
 import threading
 import time
-from typing import Any, Optional
+from typing import Optional
 
 from .api import serde
 from .api import disaggregatedset as dsapi
 from .api import leaderworkerset as lwsapi
 from .api.disaggregatedset import DisaggregatedSet, DisaggregatedSetRoleScaler
 from .api.leaderworkerset import LeaderWorkerSet
-from .cluster.store import ApiError, NotFoundError, Store
+from .cluster.store import ApiError, Store
 
 KIND_MODELS = {
     "leaderworkersets": (lwsapi.KIND, LeaderWorkerSet),

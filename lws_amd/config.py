@@ -10,7 +10,6 @@ address, node inventory, and the gang-scheduling provider.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Optional
 
 import yaml
 

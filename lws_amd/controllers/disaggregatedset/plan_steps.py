@@ -12,8 +12,8 @@ import argparse
 import json
 import sys
 
-from .planner import (RollingUpdateConfig, compute_all_steps,
-                      compute_total_steps, default_rolling_update_config)
+from .planner import (compute_all_steps, compute_total_steps,
+                      default_rolling_update_config)
 
 
 def main(argv=None) -> int:

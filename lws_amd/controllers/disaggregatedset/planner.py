@@ -17,7 +17,7 @@ one role at 0 while others still serve); all roles stay proportional.
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

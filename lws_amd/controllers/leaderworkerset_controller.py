@@ -11,16 +11,15 @@ from typing import Optional
 
 from ..api import leaderworkerset as lwsapi
 from ..api import serde
-from ..api.core import (Service, ServiceSpec, StatefulSet, StatefulSetOrdinals,
-                        StatefulSetSpec, StatefulSetUpdateStrategy,
-                        RollingUpdateStatefulSetStrategy, PodTemplateSpec)
+from ..api.core import (RollingUpdateStatefulSetStrategy, Service,
+                        ServiceSpec, StatefulSet, StatefulSetSpec,
+                        StatefulSetUpdateStrategy)
 from ..api.leaderworkerset import LeaderWorkerSet
 from ..api.meta import (LabelSelector, OwnerReference, format_label_selector,
                         get_int_or_percent, new_condition)
 from ..cluster.controller import Controller, Manager
-from ..cluster.statefulset_controller import (parse_parent_and_ordinal,
-                                              statefulset_ready)
-from ..cluster.store import AlreadyExistsError, ConflictError, Store
+from ..cluster.statefulset_controller import statefulset_ready
+from ..cluster.store import AlreadyExistsError, Store
 from ..utils import revision as revisionutils
 from ..utils.podutils import pod_running_and_ready
 

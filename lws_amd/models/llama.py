@@ -13,9 +13,7 @@ weights are always random-init — there is no network for checkpoints).
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
-from typing import Optional
-
+from dataclasses import dataclass
 import torch
 
 from ..parallel import tp as tpmod

@@ -9,8 +9,6 @@ references for tests and for CPU-only development.
 """
 from __future__ import annotations
 
-import ctypes
-import math
 from pathlib import Path
 from typing import Optional
 

@@ -6,8 +6,6 @@ as a GPU serving path.
 """
 from __future__ import annotations
 
-import math
-
 import torch
 
 

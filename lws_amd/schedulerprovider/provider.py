@@ -13,7 +13,7 @@ from __future__ import annotations
 from typing import Optional
 
 from ..api import leaderworkerset as lwsapi
-from ..api.core import Pod, PodGroup, PodGroupSpec, pod_requests_amd_gpus
+from ..api.core import Pod, PodGroup, PodGroupSpec
 from ..api.meta import IntOrString, OwnerReference
 from ..cluster.scheduler import POD_GROUP_ANNOTATION
 from ..cluster.store import AlreadyExistsError, Store
