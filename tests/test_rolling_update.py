@@ -294,3 +294,77 @@ def test_exclusive_topology_placement():
         assert len(set(all_islands)) == 4
     finally:
         c.stop()
+
+
+def test_get_int_or_percent_math():
+    """intstr.GetScaledValueFromIntOrPercent parity (reference uses round-up
+    for maxSurge, round-down for maxUnavailable — utils.go usage)."""
+    from lws_amd.api.meta import get_int_or_percent, is_percent
+    assert get_int_or_percent("50%", 4, False) == 2
+    assert get_int_or_percent("50%", 5, False) == 2   # round down
+    assert get_int_or_percent("50%", 5, True) == 3    # round up
+    assert get_int_or_percent("25%", 4, True) == 1
+    assert get_int_or_percent("10%", 4, False) == 0
+    assert get_int_or_percent("100%", 7, True) == 7
+    assert get_int_or_percent(3, 4, False) == 3
+    assert get_int_or_percent(None, 4, True) == 0
+    assert is_percent("30%") and not is_percent(3) and not is_percent("3")
+
+
+def test_rolling_update_with_percent_surge_and_unavailable(cluster):
+    """Percent-based maxUnavailable/maxSurge drive the same burst+reclaim
+    machinery as integers (reference rollingUpdateParameters resolves both
+    via intstr against Spec.Replicas)."""
+    from lws_amd.api.leaderworkerset import (RollingUpdateConfiguration,
+                                             RolloutStrategy)
+    from tests.conftest import retry_update
+
+    lws = make_lws(name="pct", replicas=4, size=2)
+    # 50% of 4 → maxUnavailable 2 (round down); 25% of 4 → maxSurge 1 (round up)
+    lws.spec.rollout_strategy = RolloutStrategy(
+        type="RollingUpdate",
+        rolling_update_configuration=RollingUpdateConfiguration(
+            partition=0, max_unavailable="50%", max_surge="25%"))
+    cluster.store.create(lws)
+    _wait_available(cluster, "pct")
+
+    def set_image(o):
+        o.spec.leader_worker_template.worker_template.spec.containers[0] \
+            .image = "engine:v2"
+    retry_update(cluster.store, "LeaderWorkerSet", "default", "pct", set_image)
+
+    saw_surge = {"max": 0}
+
+    def done():
+        sts = cluster.store.try_get("StatefulSet", "default", "pct")
+        if sts is not None:
+            saw_surge["max"] = max(saw_surge["max"], sts.spec.replicas)
+        c = cluster.get_lws("default", "pct")
+        cond = lws_condition(c, "Available")
+        live = [p for p in cluster.store.list("Pod", "default")
+                if p.metadata.deletion_timestamp is None]
+        return (c if cond is not None and cond.status == "True"
+                and c.status.updated_replicas == 4
+                and sts is not None and sts.spec.replicas == 4
+                and all(p.spec.containers[0].image == "engine:v2"
+                        for p in live) else None)
+    wait_for(done, desc="percent rollout complete + reclaimed", timeout=60)
+    # surge burst: replicas + maxSurge(25% of 4 → 1) = 5, reclaimed to 4
+    assert saw_surge["max"] == 5, saw_surge
+
+
+def test_percent_zero_budget_rejected():
+    """Webhook parity: percent values that resolve to 0/0 are invalid
+    (validation_test.go — maxUnavailable must not be 0 when maxSurge is 0)."""
+    from lws_amd.api.leaderworkerset import (RollingUpdateConfiguration,
+                                             RolloutStrategy)
+    from lws_amd.cluster.store import InvalidError
+    from lws_amd.webhooks.leaderworkerset_webhook import validate_lws
+
+    lws = make_lws(name="pct0", replicas=4, size=2)
+    lws.spec.rollout_strategy = RolloutStrategy(
+        type="RollingUpdate",
+        rolling_update_configuration=RollingUpdateConfiguration(
+            partition=0, max_unavailable="10%", max_surge="0%"))
+    with pytest.raises(InvalidError):
+        validate_lws(lws, None)
