@@ -168,3 +168,46 @@ def test_recreate_group_after_start_gate():
         wait_for(recreated, desc="group recreated after start", timeout=30)
     finally:
         c.stop()
+
+
+def test_all_example_manifests_parse_and_validate():
+    """Every YAML under examples/ round-trips through serde and passes the
+    matching webhook's defaulting + validation (reference config/samples)."""
+    import os
+
+    import yaml as pyyaml
+
+    from lws_amd.api import serde
+    from lws_amd.api.disaggregatedset import DisaggregatedSet
+    from lws_amd.api.leaderworkerset import LeaderWorkerSet
+    from lws_amd.webhooks.disaggregatedset_webhook import validate_ds
+    from lws_amd.webhooks.leaderworkerset_webhook import (default_lws,
+                                                          validate_lws)
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exdir = os.path.join(repo, "examples")
+    seen = 0
+    for fn in sorted(os.listdir(exdir)):
+        if not fn.endswith(".yaml"):
+            continue
+        with open(os.path.join(exdir, fn)) as f:
+            for doc in pyyaml.safe_load_all(f):
+                if not doc or "kind" not in doc:
+                    continue
+                kind = doc["kind"]
+                if kind == "LeaderWorkerSet":
+                    obj = serde.from_dict(LeaderWorkerSet, doc)
+                    default_lws(obj)
+                    validate_lws(obj, None)
+                    rt = serde.from_dict(LeaderWorkerSet, serde.to_dict(obj))
+                    assert rt.spec.leader_worker_template.size == \
+                        obj.spec.leader_worker_template.size
+                    seen += 1
+                elif kind == "DisaggregatedSet":
+                    obj = serde.from_dict(DisaggregatedSet, doc)
+                    validate_ds(obj, None)
+                    rt = serde.from_dict(DisaggregatedSet, serde.to_dict(obj))
+                    assert [r.name for r in rt.spec.roles] == \
+                        [r.name for r in obj.spec.roles]
+                    seen += 1
+    assert seen >= 3, f"expected >=3 example CRs, saw {seen}"
