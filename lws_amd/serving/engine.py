@@ -34,12 +34,35 @@ class EngineConfig:
 
 
 @dataclass
+class SamplingParams:
+    """Per-request sampling controls (vLLM SamplingParams subset).
+
+    temperature == 0 is greedy argmax (the default, and the only mode the
+    bench uses so measured numbers never depend on RNG).  top_k/top_p
+    restrict the nucleus before multinomial sampling; a seed makes the
+    request's stream deterministic.
+    """
+
+    temperature: float = 0.0
+    top_p: float = 1.0
+    top_k: int = 0                 # 0 = disabled
+    seed: Optional[int] = None
+    stop_token: Optional[int] = None
+
+    @property
+    def greedy(self) -> bool:
+        return self.temperature <= 0.0
+
+
+@dataclass
 class Sequence:
     seq_id: int
     token_ids: list[int]
     pages: list[int] = field(default_factory=list)
     num_cached: int = 0            # tokens already in the KV cache
     finished: bool = False
+    sampling: SamplingParams = field(default_factory=SamplingParams)
+    generator: Optional[torch.Generator] = None
 
     def __len__(self) -> int:
         return len(self.token_ids)
@@ -153,10 +176,16 @@ class Engine:
             torch.cuda.empty_cache()
 
     # -- request management --------------------------------------------
-    def add_request(self, prompt_ids: list[int]) -> int:
+    def add_request(self, prompt_ids: list[int],
+                    sampling: Optional[SamplingParams] = None) -> int:
         sid = self._next_seq_id
         self._next_seq_id += 1
-        self.sequences[sid] = Sequence(seq_id=sid, token_ids=list(prompt_ids))
+        sp = sampling or SamplingParams()
+        seq = Sequence(seq_id=sid, token_ids=list(prompt_ids), sampling=sp)
+        if sp.seed is not None:
+            seq.generator = torch.Generator(device=self.device)
+            seq.generator.manual_seed(sp.seed)
+        self.sequences[sid] = seq
         return sid
 
     def finish(self, seq_id: int) -> None:
@@ -187,6 +216,40 @@ class Engine:
             return self._step_decode(decode[:self.cfg.max_batch])
         return {}
 
+    def _sample(self, logits: torch.Tensor,
+                seqs: list[Sequence]) -> torch.Tensor:
+        """Per-sequence sampling over [len(seqs), vocab] logits.
+
+        Greedy rows stay vectorized argmax; sampled rows run the
+        temperature -> top-k -> top-p -> multinomial chain row-wise (the
+        serving path; the bench is all-greedy and never enters it).
+        """
+        out = logits.argmax(dim=-1)
+        for i, s in enumerate(seqs):
+            sp = s.sampling
+            if sp.greedy:
+                continue
+            row = logits[i].float() / sp.temperature
+            if 0 < sp.top_k < row.numel():
+                kth = torch.topk(row, sp.top_k).values[-1]
+                row = torch.where(row < kth, float("-inf"), row)
+            probs = torch.softmax(row, dim=-1)
+            if sp.top_p < 1.0:
+                sorted_p, idx = torch.sort(probs, descending=True)
+                cum = torch.cumsum(sorted_p, dim=0)
+                sorted_p[cum - sorted_p > sp.top_p] = 0.0
+                sorted_p /= sorted_p.sum()
+                pick = torch.multinomial(sorted_p, 1, generator=s.generator)
+                out[i] = idx[pick]
+            else:
+                out[i] = torch.multinomial(probs, 1, generator=s.generator)
+        return out
+
+    def _append_token(self, s: Sequence, tok: int) -> None:
+        s.token_ids.append(tok)
+        if s.sampling.stop_token is not None and tok == s.sampling.stop_token:
+            s.finished = True
+
     def _step_prefill(self, seqs: list[Sequence]) -> dict[int, int]:
         ids, pos, slots, starts = [], [], [], [0]
         for s in seqs:
@@ -209,12 +272,12 @@ class Engine:
                                       dtype=torch.int64))
         hidden = self.model.forward_prefill(batch, self.kv_caches)
         logits = self.model.compute_logits(hidden)
-        next_tokens = logits.argmax(dim=-1)
+        next_tokens = self._sample(logits, seqs)
         out = {}
         for i, s in enumerate(seqs):
             s.num_cached = len(s.token_ids)
             tok = int(next_tokens[i])
-            s.token_ids.append(tok)
+            self._append_token(s, tok)
             out[s.seq_id] = tok
         return out
 
@@ -232,8 +295,12 @@ class Engine:
             slots.append(s.pages[t // PAGE_SIZE] * PAGE_SIZE + t % PAGE_SIZE)
             lens.append(t + 1)
             bt[i, :len(s.pages)] = torch.tensor(s.pages, dtype=torch.int32)
+        all_greedy = all(s.sampling.greedy for s in seqs)
         if self.use_graphs:
-            next_tokens = self._decode_graphed(ids, pos, bt, lens, slots)
+            next_tokens, glogits = self._decode_graphed(ids, pos, bt, lens,
+                                                        slots)
+            if not all_greedy:
+                next_tokens = self._sample(glogits, seqs)
         else:
             batch = DecodeBatch(
                 input_ids=torch.tensor(ids, device=self.device,
@@ -247,12 +314,13 @@ class Engine:
                                           dtype=torch.int64))
             hidden = self.model.forward_decode(batch, self.kv_caches)
             logits = self.model.compute_logits(hidden)
-            next_tokens = logits.argmax(dim=-1)
+            next_tokens = (logits.argmax(dim=-1) if all_greedy
+                           else self._sample(logits, seqs))
         out = {}
         for i, s in enumerate(seqs):
             tok = int(next_tokens[i])
             s.num_cached = len(s.token_ids)
-            s.token_ids.append(tok)
+            self._append_token(s, tok)
             out[s.seq_id] = tok
         return out
 
@@ -271,7 +339,8 @@ class Engine:
         if g is None:
             g = _CapturedDecode(self, b_bucket, p_bucket)
             self._graphs[key] = g
-        return g.run(ids, pos, bt, lens, slots)[:B]
+        tokens = g.run(ids, pos, bt, lens, slots)
+        return tokens[:B], g.logits[:B]
 
     # -- convenience ----------------------------------------------------
     def generate(self, prompts: list[list[int]],
@@ -308,6 +377,10 @@ class _CapturedDecode:
         self.lens = torch.ones(B, dtype=torch.int32, device=dev)
         self.slots = torch.zeros(B, dtype=torch.int64, device=dev)
         self.tokens = torch.zeros(B, dtype=torch.long, device=dev)
+        # static logits output so sampled requests can re-sample outside
+        # the captured graph (greedy argmax happens inside it)
+        self.logits = torch.zeros(B, engine.model_cfg.vocab_size,
+                                  dtype=torch.float32, device=dev)
         self.batch = DecodeBatch(input_ids=self.ids, positions=self.pos,
                                  block_tables=self.bt, seq_lens=self.lens,
                                  slot_mapping=self.slots)
@@ -318,6 +391,7 @@ class _CapturedDecode:
         eng = self.engine
         hidden = eng.model.forward_decode(self.batch, eng.kv_caches)
         logits = eng.model.compute_logits(hidden)
+        self.logits.copy_(logits)
         torch.argmax(logits, dim=-1, out=self.tokens)
 
     def _capture(self) -> None:

@@ -21,7 +21,7 @@ from concurrent.futures import Future
 from dataclasses import dataclass, field
 from typing import Optional
 
-from .engine import Engine, EngineConfig
+from .engine import Engine, EngineConfig, SamplingParams
 
 
 @dataclass
@@ -49,9 +49,10 @@ class ServingLoop:
         self._thread.start()
         return self
 
-    def submit(self, prompt_ids: list[int], max_tokens: int) -> Future:
+    def submit(self, prompt_ids: list[int], max_tokens: int,
+               sampling: Optional[SamplingParams] = None) -> Future:
         with self._lock:
-            sid = self.engine.add_request(list(prompt_ids))
+            sid = self.engine.add_request(list(prompt_ids), sampling)
             p = _Pending(seq_id=sid, prompt_len=len(prompt_ids),
                          max_tokens=max_tokens)
             self._pending[sid] = p
@@ -84,7 +85,7 @@ class ServingLoop:
                             finished.append(sid)
                             continue
                         produced = len(seq.token_ids) - p.prompt_len
-                        if produced >= p.max_tokens:
+                        if produced >= p.max_tokens or seq.finished:
                             tokens = seq.token_ids[
                                 p.prompt_len:p.prompt_len + p.max_tokens]
                             self.engine.finish(sid)
@@ -142,7 +143,12 @@ def build_app(loop: ServingLoop, model_name: str):
             vocab = loop.engine.model_cfg.vocab_size
             prompt = [b % vocab for b in prompt.encode()] or [0]
         max_tokens = int(body.get("max_tokens", 16))
-        fut = loop.submit(prompt, max_tokens)
+        sp = SamplingParams(
+            temperature=float(body.get("temperature", 0.0)),
+            top_p=float(body.get("top_p", 1.0)),
+            top_k=int(body.get("top_k", 0)),
+            seed=(int(body["seed"]) if "seed" in body else None))
+        fut = loop.submit(prompt, max_tokens, sp)
         tokens = fut.result(timeout=float(body.get("timeout", 300)))
         return {
             "id": f"cmpl-{uuid.uuid4().hex[:12]}",
