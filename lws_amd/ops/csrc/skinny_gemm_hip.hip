@@ -70,6 +70,36 @@ __device__ __forceinline__ void sg_stage_async(
   }
 }
 
+using i32x4_t = __attribute__((ext_vector_type(4))) int;
+
+// ds_read_b128 via inline asm.  The LLVM waitcnt pass cannot prove our
+// counted vmcnt waits cover the global_load_lds writes (the read aliases
+// every pipeline buffer), so a compiled LDS read gets a forced vmcnt(0)
+// inserted before it — draining the whole pipeline each iteration.  An
+// asm read is invisible to the pass; the explicit s_waitcnt lgkmcnt(0)
+// before first use is the caller's job (guide §5 template / m203).
+__device__ __forceinline__ i32x4_t sg_ds_read_b128(const ushort* lds,
+                                                   int byte_off) {
+  typedef __attribute__((address_space(3))) const int lds_c32;
+  lds_c32* addr = (lds_c32*)(reinterpret_cast<const char*>(lds) + byte_off);
+  i32x4_t r;
+  asm volatile("ds_read_b128 %0, %1" : "=v"(r) : "v"(addr));
+  return r;
+}
+
+// s_waitcnt vmcnt(N) with a literal count — the counted-vmcnt idiom
+// (guide §5 K-loop): wait until at most N vmem ops are outstanding, so
+// staging for future sub-slices stays in flight across the barrier
+// instead of draining to 0 every iteration.
+template <int N>
+__device__ __forceinline__ void sg_wait_vm() {
+  if constexpr (N == 0)  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if constexpr (N == 5)  asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+  if constexpr (N == 6)  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  if constexpr (N == 10) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+  if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+}
+
 template <int MTILES, bool XLDS>
 __global__ __launch_bounds__(256)
 void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
@@ -85,8 +115,18 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
   const int frag_row = lane % 16;        // m (A) / n (B)
   const int frag_kgrp = lane / 16;       // which 8-wide k group
 
-  __shared__ ushort w_lds[2][SG_ROWS * SG_KSUB];
-  __shared__ ushort x_lds[XLDS ? 2 : 1][XLDS ? 16 * MTILES * SG_KSUB : 1];
+  // XLDS: 3-deep pipeline, 2 sub-slices staged ahead, counted-vmcnt waits
+  // (all in-loop vmem is staging, so vmcnt counts are exact).  3 buffers
+  // keep LDS at 72 KB -> 2 workgroups resident per CU, doubling in-flight
+  // staging in the latency-bound cold-weights regime.  !XLDS: the in-loop
+  // global x reads pollute vmcnt, so keep the 2-phase full drain.
+  constexpr int NBUF = XLDS ? 3 : 2;
+  // global_load_lds instructions per wave per sub-slice (1 KiB each):
+  // W 16 KiB -> 4/wave; x MTILES*4 KiB -> MTILES/wave
+  constexpr int LOADS = 4 + (XLDS ? MTILES : 0);
+
+  __shared__ ushort w_lds[NBUF][SG_ROWS * SG_KSUB];
+  __shared__ ushort x_lds[XLDS ? NBUF : 1][XLDS ? 16 * MTILES * SG_KSUB : 1];
 
   f32x4_t acc[MTILES];
 #pragma unroll
@@ -96,67 +136,91 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
   const int w_rows_valid = min(SG_ROWS, N - n0);
   const int nsub = (kend - kbegin) / SG_KSUB;
 
-  // prologue: stage sub-slice 0 into buffer 0
-  sg_stage_async(w_lds[0], w_base + kbegin, K, SG_ROWS, w_rows_valid, wave,
-                 lane);
-  if (XLDS)
-    sg_stage_async(x_lds[0], x + kbegin, K, 16 * MTILES, M, wave, lane);
-  __syncthreads();
+  auto stage = [&](int s) {
+    const int ks = kbegin + s * SG_KSUB;
+    sg_stage_async(w_lds[s % NBUF], w_base + ks, K, SG_ROWS, w_rows_valid,
+                   wave, lane);
+    if (XLDS)
+      sg_stage_async(x_lds[XLDS ? s % NBUF : 0], x + ks, K, 16 * MTILES, M,
+                     wave, lane);
+  };
+
+  // prologue: stage sub-slices 0..NBUF-2 (XLDS: 3 in flight, 2-phase: 1)
+  for (int s = 0; s < min(nsub, NBUF - 1); ++s) stage(s);
+  if (!XLDS) __syncthreads();
 
   for (int s = 0; s < nsub; ++s) {
-    const int cur = s & 1;
-    if (s + 1 < nsub) {
-      const int ks_next = kbegin + (s + 1) * SG_KSUB;
-      sg_stage_async(w_lds[cur ^ 1], w_base + ks_next, K, SG_ROWS,
-                     w_rows_valid, wave, lane);
-      if (XLDS)
-        sg_stage_async(x_lds[cur ^ 1], x + ks_next, K, 16 * MTILES, M, wave,
-                       lane);
+    const int cur = s % NBUF;
+    if (XLDS) {
+      // own loads for sub-slice s are complete once at most
+      // LOADS x (stages issued after s, i.e. s+1) remain outstanding
+      const int ahead = min(NBUF - 2, nsub - 1 - s);
+      if (ahead >= 1) sg_wait_vm<LOADS>();
+      else            sg_wait_vm<0>();
+      // raw barrier (no implicit vmcnt(0) drain): after it, every wave's
+      // stage of s has landed and every wave is done reading buffer s-1,
+      // which stage(s+2) below overwrites
+      __builtin_amdgcn_s_barrier();
+      if (s + NBUF - 1 < nsub) stage(s + NBUF - 1);
+    } else if (s + 1 < nsub) {
+      stage(s + 1);
     }
     // compute current sub-slice: 4 k-steps of 32
     const ushort* wt = w_lds[cur];
-    const ushort* xt = XLDS ? x_lds[cur & (XLDS ? 1 : 0)] : nullptr;
+    const ushort* xt = XLDS ? x_lds[XLDS ? cur : 0] : nullptr;
 #pragma unroll
     for (int k0 = 0; k0 < SG_KSUB; k0 += 32) {
       const int colb = (k0 + frag_kgrp * 8) * 2;
-      bf16x8_t bfrag;
-      {
-        const int row = wave * SG_NTILE + frag_row;
-        bf16x8 tmp;
-        tmp.u = *reinterpret_cast<const uint4*>(
-            reinterpret_cast<const char*>(wt) + row * SG_ROWB +
-            sg_swz(row, colb));
-#pragma unroll
-        for (int j = 0; j < 8; ++j) bfrag[j] = (short)tmp.h[j];
-      }
+      const int brow = wave * SG_NTILE + frag_row;
+      i32x4_t braw = sg_ds_read_b128(wt, brow * SG_ROWB + sg_swz(brow, colb));
+      i32x4_t araw[MTILES];
 #pragma unroll
       for (int t = 0; t < MTILES; ++t) {
         const int m = t * 16 + frag_row;
-        bf16x8_t afrag;
-        {
+        if (XLDS) {
+          araw[t] = sg_ds_read_b128(xt, m * SG_ROWB + sg_swz(m, colb));
+        } else {
+          // x is tiny and L2-resident; read the fragment from global
+          const int mm = m < M ? m : M - 1;
           bf16x8 tmp;
-          if (XLDS) {
-            tmp.u = *reinterpret_cast<const uint4*>(
-                reinterpret_cast<const char*>(xt) + m * SG_ROWB +
-                sg_swz(m, colb));
-          } else {
-            // x is tiny and L2-resident; read the fragment from global
-            const int mm = m < M ? m : M - 1;
-            tmp.u = *reinterpret_cast<const uint4*>(
-                x + (long long)mm * K + kbegin + s * SG_KSUB + k0
-                + frag_kgrp * 8);
-            if (m >= M) tmp.u = make_uint4(0, 0, 0, 0);
-          }
-#pragma unroll
-          for (int j = 0; j < 8; ++j) afrag[j] = (short)tmp.h[j];
+          tmp.u = *reinterpret_cast<const uint4*>(
+              x + (long long)mm * K + kbegin + s * SG_KSUB + k0
+              + frag_kgrp * 8);
+          if (m >= M) tmp.u = make_uint4(0, 0, 0, 0);
+          __builtin_memcpy(&araw[t], &tmp.u, 16);
         }
+      }
+      // the fragment registers are threaded through the wait as "+v"
+      // operands: without that the scheduler hoists MFMAs (data-dependent
+      // only on the asm ds_reads) above the wait — a load/use race, since
+      // CDNA has no hardware interlock on LDS reads
+      if constexpr (MTILES == 1)
+        asm volatile("s_waitcnt lgkmcnt(0)"
+                     : "+v"(braw), "+v"(araw[0])::"memory");
+      else
+        asm volatile("s_waitcnt lgkmcnt(0)"
+                     : "+v"(braw), "+v"(araw[0]), "+v"(araw[1])::"memory");
+      bf16x8 bw;
+      __builtin_memcpy(&bw.u, &braw, 16);
+      bf16x8_t bfrag;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) bfrag[j] = (short)bw.h[j];
+#pragma unroll
+      for (int t = 0; t < MTILES; ++t) {
+        bf16x8 aw;
+        __builtin_memcpy(&aw.u, &araw[t], 16);
+        bf16x8_t afrag;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) afrag[j] = (short)aw.h[j];
         acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[t],
                                                          0, 0, 0);
       }
     }
-    // drains the in-flight global_load_lds of the next buffer AND fences
-    // the current buffer for reuse (minimum 2-phase schedule)
-    __syncthreads();
+    if (!XLDS) {
+      // 2-phase drain: __syncthreads' implicit vmcnt(0) completes the
+      // next buffer's staging AND fences the current buffer for reuse
+      __syncthreads();
+    }
   }
 
   // C/D layout: lane l reg r -> D[row=(l/16)*4+r][col=l%16].
@@ -212,7 +276,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const int n_blocks = (N + SG_ROWS - 1) / SG_ROWS;
   // split K so the grid lands near the target workgroup count —
   // balances chip fill against split-K workspace traffic
-  const int target = env_int("LWS_SG_TARGET", 1024);
+  const int target = env_int("LWS_SG_TARGET", 256);
   int split = target / max(1, n_blocks);
   const int max_split = max(1, K / SG_KSUB);
   if (split > max_split) split = max_split;

@@ -106,12 +106,13 @@ class ShardedLinear:
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
         if x.is_cuda and x.dim() == 2 and x.size(0) <= 32 and \
-                x.size(1) % 128 == 0 and (self.weight.numel() <= 32 * 1024 * 1024
-                                          or self.weight.size(0) >= 65536):
+                x.size(1) % 128 == 0 and self.weight.numel() <= 32 * 1024 * 1024:
             # decode-shape path for small weight shards (<=64 MB bf16, the
-            # TP-sharded regime): hand-written split-K MFMA kernel — the
-            # hipBLASLt heuristic underfills the chip there (see profiles/).
-            # Large shards stay on hipBLASLt, which fills the chip fine.
+            # TP-sharded regime): the split-K counted-vmcnt MFMA kernel
+            # beats the hipBLASLt heuristic cold AND warm there (event-
+            # timed A/B in profiles/r01_skinny_dispatch.md).  Bigger
+            # shards (gate_up, lm_head) stay on hipBLASLt, which wins
+            # once its tiles fill the chip.
             from .. import ops
             return ops.skinny_gemm(x.contiguous(), self.weight)
         return x @ self.weight.t()
