@@ -117,16 +117,21 @@ def paged_attention_decode(q: torch.Tensor, k_cache: torch.Tensor,
     G = Hq // Hkv
     max_len = int(block_tables.size(1)) * int(k_cache.size(2))
     if chunk_keys is None:
-        # flash-decoding split: size chunks so the launch approaches ~1024
-        # workgroups (4 waves each) and 256 CUs stay busy even at B*Hkv=1
-        target_chunks = max(1, 1024 // max(1, B * Hkv))
-        target_chunks = min(target_chunks, (max_len + 63) // 64)
-        chunk_keys = -(-max_len // target_chunks)
-        if max_len >= 1024:
-            # long contexts: per-chunk fixed costs (q staging, wave merge)
-            # dominate below 128 keys (measured A/B in profiles/)
-            chunk_keys = max(chunk_keys, 128)
-        chunk_keys = ((chunk_keys + 15) // 16) * 16
+        if B * Hkv >= 256 and G == 1 and max_len <= 512:
+            # chip already full at one block per (b, hkv): single chunk
+            # takes the in-kernel-normalize DIRECT path (no ws, no reduce
+            # launch) — wins only at G=1 (profiles/r01_pa_direct.md)
+            chunk_keys = ((max_len + 15) // 16) * 16
+        else:
+            # flash-decoding split (profiles/r01_pa_direct.md chunk sweep):
+            # the optimum across B*Hkv in {32, 256} x len 160..4096 is
+            # ~16 chunks per sequence when B*Hkv is small (fills the chip)
+            # and ~8 when one block per (b, hkv) already covers it —
+            # bounded below by 32-key chunks (per-chunk fixed costs)
+            target_chunks = 16 if B * Hkv <= 64 else 8
+            target_chunks = min(target_chunks, (max_len + 31) // 32)
+            chunk_keys = -(-max_len // target_chunks)
+            chunk_keys = ((chunk_keys + 15) // 16) * 16
     num_chunks = max(1, (max_len + chunk_keys - 1) // chunk_keys)
     if workspace is None:
         ws_acc = torch.empty((B, Hkv, num_chunks, G, D), dtype=torch.float32,
@@ -164,7 +169,7 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
     if out is None:
         out = torch.empty(M, N, dtype=x.dtype, device=x.device)
     import os
-    target = int(os.environ.get("LWS_SG_TARGET", "1024"))
+    target = int(os.environ.get("LWS_SG_TARGET", "256"))  # keep = C++ default
     n_blocks = (N + 63) // 64
     split = min(max(1, target // max(1, n_blocks)), max(1, K // 128))
     k_slice = (K // split + 127) // 128 * 128

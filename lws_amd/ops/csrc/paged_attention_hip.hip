@@ -27,9 +27,18 @@
 // G is a template parameter so the per-head loops fully unroll and the
 // accumulator arrays stay in VGPRs — runtime-indexed register arrays are
 // demoted to scratch memory (guide rule #20), which costs ~25x here.
-template <int G>
-__global__ __launch_bounds__(256)
+// DIRECT: single-chunk fast path — normalize by l in the epilogue and
+// write bf16 straight to out, skipping the ws round-trip AND the reduce
+// kernel launch (pays off when B*Hkv alone fills the chip — measured in
+// profiles/r01_pa_direct.md).
+// NW: waves per block.  NW=1 skips the inter-wave LDS merge entirely and
+// lets the scheduler pack many 64-thread blocks per CU — the latency-
+// hiding mode for the TP8 B*Hkv=32 regime where 4-wave blocks leave the
+// chip 1-wave-per-SIMD.
+template <int G, bool DIRECT, int NW>
+__global__ __launch_bounds__(NW * WAVE_SIZE)
 void paged_attention_chunk_kernel(
+    ushort* __restrict__ out,        // [B, Hq, 128]   (DIRECT only)
     float* __restrict__ ws_acc,      // [B, Hkv, chunks, G, 128]
     float* __restrict__ ws_ml,       // [B, Hkv, chunks, G, 2]  (m, l)
     const ushort* __restrict__ q,    // [B, Hq, 128] (row stride q_stride)
@@ -99,11 +108,11 @@ void paged_attention_chunk_kernel(
   const long long kv_page_stride = (long long)Hkv * page_size * PA_HEAD_DIM;
 
   // wave-interleaved key streams: wave w takes keys [kstart+w*4+kgrp],
-  // stepping 16 keys per workgroup pass — no block-level sync inside.
+  // stepping 4*NW keys per workgroup pass — no block-level sync inside.
   // unroll 2 so the next pass's page-table + K/V loads issue under the
   // current pass's softmax (latency cover for the 1-wave/SIMD TP8 regime)
 #pragma unroll 2
-  for (int k0 = kstart + wave * 4; k0 < kend; k0 += 4 * PA_NWAVES) {
+  for (int k0 = kstart + wave * 4; k0 < kend; k0 += 4 * NW) {
     const int key = k0 + kgrp;
     const bool valid = key < kend;
     float kf[8];
@@ -171,12 +180,51 @@ void paged_attention_chunk_kernel(
     l_run[g] = l4;
   }
 
-  // merge the 4 waves' (m, l, acc) through LDS with softmax rescaling.
+  const long long wsbase =
+      (((long long)b * Hkv + hkv) * num_chunks + chunk) * G * PA_HEAD_DIM;
+
+  if (NW == 1) {
+    // single-wave block: no inter-wave merge.  After the cross-group
+    // shfl sums every lane holds the final acc for its slice; the
+    // kgrp==0 lane quarter writes it out.
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v2 = acc[g][j];
+        v2 += __shfl_xor(v2, 16, WAVE_SIZE);
+        v2 += __shfl_xor(v2, 32, WAVE_SIZE);
+        acc[g][j] = v2;
+      }
+      if (lane < PA_SLICES) {
+        if (DIRECT) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            out[(((long long)b * Hkv + hkv) * G + g) * PA_HEAD_DIM
+                + slice * 8 + j] =
+                f32_to_bf16(acc[g][j] / fmaxf(l_run[g], 1e-20f));
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            ws_acc[wsbase + (long long)g * PA_HEAD_DIM + slice * 8 + j] =
+                acc[g][j];
+          if (lane == 0) {
+            const long long mlbase =
+                ((((long long)b * Hkv + hkv) * num_chunks + chunk) * G + g)
+                * 2;
+            ws_ml[mlbase] = m_run[g];
+            ws_ml[mlbase + 1] = l_run[g];
+          }
+        }
+      }
+    }
+    return;
+  }
+
+  // merge the NW waves' (m, l, acc) through LDS with softmax rescaling.
   __shared__ float mw[PA_MAX_GQA][PA_NWAVES];
   __shared__ float lw[PA_MAX_GQA][PA_NWAVES];
   __shared__ float aw[PA_NWAVES][PA_SLICES][8];
-  const long long wsbase =
-      (((long long)b * Hkv + hkv) * num_chunks + chunk) * G * PA_HEAD_DIM;
   for (int g = 0; g < G; ++g) {
     // within-wave: sum acc over the 4 key groups (lane bits 4,5)
 #pragma unroll
@@ -207,12 +255,17 @@ void paged_attention_chunk_kernel(
         accv += aw[w][s][j] * wgt;
         lv += lw[g][w] * wgt;
       }
-      ws_acc[wsbase + (long long)g * PA_HEAD_DIM + s * 8 + j] = accv;
-      if (s == 0 && j == 0) {
-        const long long mlbase =
-            ((((long long)b * Hkv + hkv) * num_chunks + chunk) * G + g) * 2;
-        ws_ml[mlbase] = m_g;
-        ws_ml[mlbase + 1] = lv;
+      if (DIRECT) {
+        out[(((long long)b * Hkv + hkv) * G + g) * PA_HEAD_DIM + s * 8 + j] =
+            f32_to_bf16(accv / fmaxf(lv, 1e-20f));
+      } else {
+        ws_acc[wsbase + (long long)g * PA_HEAD_DIM + s * 8 + j] = accv;
+        if (s == 0 && j == 0) {
+          const long long mlbase =
+              ((((long long)b * Hkv + hkv) * num_chunks + chunk) * G + g) * 2;
+          ws_ml[mlbase] = m_g;
+          ws_ml[mlbase + 1] = lv;
+        }
       }
     }
     __syncthreads();
@@ -320,29 +373,51 @@ void paged_attention_decode(torch::Tensor out, torch::Tensor q,
 
   hipStream_t stream = at::hip::getCurrentHIPStream();
   dim3 grid(B, Hkv, num_chunks);
-#define PA_LAUNCH(GG)                                                        \
-  hipLaunchKernelGGL((paged_attention_chunk_kernel<GG>), grid, dim3(256), 0,  \
-                     stream, ws_acc.data_ptr<float>(), ws_ml.data_ptr<float>(),\
+  const bool direct = (num_chunks == 1);
+  // 4-wave blocks beat 1-wave blocks at every measured decode shape
+  // (profiles/r01_pa_direct.md NW sweep); LWS_PA_NW=1 keeps the 1-wave
+  // variant reachable for tuning.
+  static const int nw_env = [] {
+    const char* v = getenv("LWS_PA_NW");
+    return v ? atoi(v) : 0;
+  }();
+  const bool wave1 = nw_env == 1;
+#define PA_LAUNCH(GG, DD, NW)                                                \
+  hipLaunchKernelGGL((paged_attention_chunk_kernel<GG, DD, NW>), grid,        \
+                     dim3(NW * WAVE_SIZE), 0,                                 \
+                     stream, (ushort*)out.data_ptr(),                         \
+                     ws_acc.data_ptr<float>(), ws_ml.data_ptr<float>(),       \
                      (const ushort*)q.data_ptr(),                             \
                      (const ushort*)k_cache.data_ptr(),                       \
                      (const ushort*)v_cache.data_ptr(),                       \
                      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),  \
                      (float)scale, Hkv, page_size, max_pages,                 \
                      (int)chunk_keys, num_chunks, (long long)q.stride(0))
+#define PA_DISPATCH(GG)                                                      \
+  do {                                                                       \
+    if (direct) {                                                            \
+      if (wave1) PA_LAUNCH(GG, true, 1); else PA_LAUNCH(GG, true, 4);         \
+    } else {                                                                 \
+      if (wave1) PA_LAUNCH(GG, false, 1); else PA_LAUNCH(GG, false, 4);       \
+    }                                                                        \
+  } while (0)
   switch (G) {
-    case 1: PA_LAUNCH(1); break;
-    case 2: PA_LAUNCH(2); break;
-    case 4: PA_LAUNCH(4); break;
-    case 8: PA_LAUNCH(8); break;
-    case 16: PA_LAUNCH(16); break;
+    case 1: PA_DISPATCH(1); break;
+    case 2: PA_DISPATCH(2); break;
+    case 4: PA_DISPATCH(4); break;
+    case 8: PA_DISPATCH(8); break;
+    case 16: PA_DISPATCH(16); break;
     default:
       TORCH_CHECK(false, "paged_attention: GQA group must be 1/2/4/8/16");
   }
+#undef PA_DISPATCH
 #undef PA_LAUNCH
-  hipLaunchKernelGGL(paged_attention_reduce_kernel, dim3(B, Hq), dim3(128), 0,
-                     stream, (ushort*)out.data_ptr(), ws_acc.data_ptr<float>(),
-                     ws_ml.data_ptr<float>(), seq_lens.data_ptr<int>(),
-                     G, Hkv, (int)chunk_keys, num_chunks);
+  if (!direct)
+    hipLaunchKernelGGL(paged_attention_reduce_kernel, dim3(B, Hq), dim3(128),
+                       0, stream, (ushort*)out.data_ptr(),
+                       ws_acc.data_ptr<float>(), ws_ml.data_ptr<float>(),
+                       seq_lens.data_ptr<int>(), G, Hkv, (int)chunk_keys,
+                       num_chunks);
 }
 
 void reshape_and_cache(torch::Tensor k, torch::Tensor v,
