@@ -70,24 +70,14 @@ def main(argv=None) -> int:
                                 "seed": args.seed}})
     engine = conductor.host.engine
 
+    from lws_amd.serving.runtime import CollectiveEngine
     from lws_amd.serving.server import ServingLoop, build_app
 
-    # NOTE: with world > 1 the engine steps are collective; the leader's
-    # serving loop must broadcast step commands.  v1 scope: HTTP serving is
-    # single-shard (TP=1) or driven externally via bench.py's conductor.
-    if world > 1:
-        print("lws-amd-engine: leader ready (collective mode); "
-              "serving frontend requires the conductor protocol",
-              flush=True)
-        try:
-            import time
-            while True:
-                time.sleep(1)
-        except KeyboardInterrupt:
-            conductor.command({"op": "exit"})
-        return 0
-
-    loop = ServingLoop(engine).start()
+    # with world > 1 the engine steps are collective: the leader serves
+    # HTTP and broadcasts every engine call through the conductor so all
+    # shards execute the identical sequence (CollectiveEngine)
+    serving_engine = CollectiveEngine(conductor) if world > 1 else engine
+    loop = ServingLoop(serving_engine).start()
     app = build_app(loop, args.model)
     import uvicorn
 

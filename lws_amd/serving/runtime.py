@@ -125,6 +125,15 @@ def execute_command(cmd: dict, host: ShardHost, sync, t_block: dict) -> Any:
         return None
     if op == "decode_bench":
         return host.decode_bench(cmd["batch"], cmd["prompt_len"], cmd["steps"])
+    if op == "add_request":
+        from .engine import SamplingParams
+        sp = SamplingParams(**cmd["sampling"]) if cmd.get("sampling") else None
+        return host.engine.add_request(cmd["prompt"], sp)
+    if op == "step":
+        return host.engine.step()
+    if op == "finish":
+        host.engine.finish(cmd["seq_id"])
+        return None
     if op == "block_begin":
         sync()
         t_block["t0"] = time.perf_counter()
@@ -240,3 +249,55 @@ class CollectiveGroupRuntime(PodRuntime):
             return rev
         except queue.Empty:
             return None
+
+
+class CollectiveEngine:
+    """Leader-side engine facade for TP > 1 serving.
+
+    Every state-mutating engine call is broadcast through the conductor so
+    all ranks execute the identical sequence — the collectives inside
+    step() (TP all-reduce, lm_head all-gather) then line up by
+    construction.  Read-only state (sequences, model_cfg, ready) comes
+    from the leader's local shard, which all ranks mirror.
+    """
+
+    def __init__(self, conductor: "Conductor"):
+        self.conductor = conductor
+
+    @property
+    def _local(self):
+        return self.conductor.host.engine
+
+    @property
+    def sequences(self):
+        return self._local.sequences
+
+    @property
+    def model_cfg(self):
+        return self._local.model_cfg
+
+    @property
+    def ready(self):
+        return self._local.ready
+
+    def add_request(self, prompt_ids, sampling=None) -> int:
+        import dataclasses
+        import random as _random
+        payload = None
+        if sampling is not None:
+            if sampling.seed is None and not sampling.greedy:
+                # unseeded sampling would draw from each rank's own RNG
+                # stream and diverge the shards — pin a shared seed here
+                sampling = dataclasses.replace(
+                    sampling, seed=_random.getrandbits(31))
+            payload = dataclasses.asdict(sampling)
+        acks = self.conductor.command({"op": "add_request",
+                                       "prompt": list(prompt_ids),
+                                       "sampling": payload})
+        return acks[0]
+
+    def step(self) -> dict:
+        return self.conductor.command({"op": "step"})[0]
+
+    def finish(self, seq_id: int) -> None:
+        self.conductor.command({"op": "finish", "seq_id": seq_id})
