@@ -105,11 +105,22 @@ MODEL_PRESETS = {
 
 @dataclass
 class PrefillBatch:
-    """Variable-length prompt batch, flattened to [T, ...]."""
+    """Variable-length prompt batch, flattened to [T, ...].
+
+    Chunked prefill (cached prefix): kv_row_idx gathers the full context
+    (prefix + chunk) rows out of the paged KV cache (flat row index
+    (page*Hkv + h)*page_size + off per (token, head)); kv_starts are B+1
+    offsets into those gathered rows and q_offsets[b] is the number of
+    cached keys before seq b's chunk (shifts the causal frontier).
+    All three None => whole-prompt prefill (attention reads the chunk's
+    own k/v directly)."""
     input_ids: torch.Tensor          # [T] int64
     positions: torch.Tensor          # [T] int32
     seq_starts: list[int]            # len B+1 prefix offsets into T
     slot_mapping: torch.Tensor       # [T] int64 cache slots
+    kv_row_idx: Optional[torch.Tensor] = None   # [Tkv * Hkv] int64
+    kv_starts: Optional[list[int]] = None       # len B+1
+    q_offsets: Optional[list[int]] = None       # len B
 
 
 @dataclass
@@ -319,7 +330,8 @@ class LlamaForCausalLM:
             q, k, v = self._qkv_views(qkv, layer, batch.positions)
             k_cache, v_cache = kv_caches[li]
             self._write_cache(k, v, k_cache, v_cache, batch.slot_mapping)
-            attn = self._prefill_attention(q, k, v, batch, layer)
+            attn = self._prefill_attention(q, k, v, batch, layer,
+                                           k_cache, v_cache)
             o = layer.o(attn)
             x = all_reduce(o)
             # MLP
@@ -330,29 +342,47 @@ class LlamaForCausalLM:
                             device=h.device, dtype=torch.long)
         return h[last]
 
-    def _prefill_attention(self, q, k, v, batch: PrefillBatch, layer):
+    def _prefill_attention(self, q, k, v, batch: PrefillBatch, layer,
+                           k_cache, v_cache):
         """Causal varlen attention over [T, H, D] views.  GPU: the in-repo
         flash-style HIP kernel (MFMA, online softmax, no S x S
-        materialization).  CPU: per-sequence fp32 reference composition."""
-        if self._ops.is_gpu:
-            return self._ops.ops.prefill_attention(
-                q, k, v, batch.seq_starts, self.scale)
+        materialization).  CPU: per-sequence fp32 reference composition.
+        Chunked prefill gathers the full context (cached prefix + chunk,
+        both already written to the paged cache this layer) and shifts
+        the causal frontier by q_offsets."""
         cfg = self.cfg
         d = cfg.head_dim
+        kv_starts = batch.seq_starts
+        q_offsets = None
+        if batch.kv_row_idx is not None:
+            k = k_cache.reshape(-1, d).index_select(
+                0, batch.kv_row_idx).view(-1, layer.hkv, d)
+            v = v_cache.reshape(-1, d).index_select(
+                0, batch.kv_row_idx).view(-1, layer.hkv, d)
+            kv_starts = batch.kv_starts
+            q_offsets = batch.q_offsets
+        if self._ops.is_gpu:
+            return self._ops.ops.prefill_attention(
+                q, k, v, batch.seq_starts, self.scale,
+                kv_starts=kv_starts, q_offsets=q_offsets)
         T = q.size(0)
         out = torch.empty(T, layer.hq * d, device=q.device, dtype=q.dtype)
         G = layer.hq // layer.hkv
+        offs = q_offsets or [0] * (len(batch.seq_starts) - 1)
         for i in range(len(batch.seq_starts) - 1):
             s0, s1 = batch.seq_starts[i], batch.seq_starts[i + 1]
-            S = s1 - s0
+            kv0, kv1 = kv_starts[i], kv_starts[i + 1]
+            S, KV, off = s1 - s0, kv1 - kv0, offs[i]
             qs = q[s0:s1].transpose(0, 1)      # [Hq, S, D]
-            ks = k[s0:s1].transpose(0, 1)
-            vs = v[s0:s1].transpose(0, 1)
+            ks = k[kv0:kv1].transpose(0, 1)
+            vs = v[kv0:kv1].transpose(0, 1)
             ks = ks.repeat_interleave(G, dim=0)
             vs = vs.repeat_interleave(G, dim=0)
             scores = (qs @ ks.transpose(-1, -2)).float() * self.scale
-            mask = torch.full((S, S), float("-inf"),
-                              device=q.device).triu_(1)
+            cols = torch.arange(KV, device=q.device)
+            rows = torch.arange(S, device=q.device)
+            mask = torch.where(cols[None, :] > off + rows[:, None],
+                               float("-inf"), 0.0)
             p = torch.softmax(scores + mask, dim=-1).to(q.dtype)
             o = (p @ vs).transpose(0, 1).reshape(S, layer.hq * d)
             out[s0:s1] = o

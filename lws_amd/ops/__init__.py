@@ -185,26 +185,38 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
 
 def prefill_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                       seq_starts: list, scale: float,
-                      out: Optional[torch.Tensor] = None) -> torch.Tensor:
+                      out: Optional[torch.Tensor] = None,
+                      kv_starts: Optional[list] = None,
+                      q_offsets: Optional[list] = None) -> torch.Tensor:
     """Flash-style causal varlen prefill attention.
 
-    q [T, Hq, D] / k,v [T, Hkv, D] (strided row views OK, D=128);
-    seq_starts: python list of B+1 prefix offsets.  Returns [T, Hq*D].
+    q [T, Hq, D] / k,v [Tkv, Hkv, D] (strided row views OK, D=128);
+    seq_starts: python list of B+1 q-row prefix offsets.  Chunked prefill:
+    kv_starts (B+1 offsets into k/v, covering cached prefix + chunk) and
+    q_offsets (cached keys before each chunk) shift the causal frontier so
+    q row i of seq b attends keys [0, q_offsets[b] + i].  Returns [T, Hq*D].
     """
     lib = require_native()
     T, Hq, D = q.shape
     device = q.device
+    B = len(seq_starts) - 1
     tile_seq, tile_q0 = [], []
-    for i in range(len(seq_starts) - 1):
+    for i in range(B):
         S = seq_starts[i + 1] - seq_starts[i]
         for q0 in range(0, S, 64):
             tile_seq.append(i)
             tile_q0.append(q0)
+    if kv_starts is None:
+        kv_starts = seq_starts
+    if q_offsets is None:
+        q_offsets = [0] * B
     if out is None:
         out = torch.empty(T, Hq * D, dtype=q.dtype, device=device)
     lib.prefill_attention(
         out, q, k, v,
         torch.tensor(tile_seq, dtype=torch.int32, device=device),
         torch.tensor(tile_q0, dtype=torch.int32, device=device),
-        torch.tensor(seq_starts, dtype=torch.int32, device=device), scale)
+        torch.tensor(seq_starts, dtype=torch.int32, device=device),
+        torch.tensor(kv_starts, dtype=torch.int32, device=device),
+        torch.tensor(q_offsets, dtype=torch.int32, device=device), scale)
     return out

@@ -22,6 +22,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
 void prefill_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                        torch::Tensor v, torch::Tensor tile_seq,
                        torch::Tensor tile_q0, torch::Tensor seq_starts,
+                       torch::Tensor kv_starts, torch::Tensor q_offs,
                        double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -38,7 +38,9 @@ void prefill_attention_kernel(
     const ushort* __restrict__ v,      // [T, Hkv, 128], row stride v_stride
     const int* __restrict__ tile_seq,  // [tiles] sequence index
     const int* __restrict__ tile_q0,   // [tiles] first q row (seq-local)
-    const int* __restrict__ seq_starts,  // [B+1] global token offsets
+    const int* __restrict__ seq_starts,  // [B+1] global q-token offsets
+    const int* __restrict__ kv_starts,   // [B+1] global k/v row offsets
+    const int* __restrict__ q_offs,      // [B] cached keys before the chunk
     float scale, int Hq, int Hkv,
     long long q_stride, long long k_stride, long long v_stride) {
   const int tile = blockIdx.x;
@@ -55,6 +57,11 @@ void prefill_attention_kernel(
   const int q0 = tile_q0[tile];              // seq-local first q row of tile
   const int q_rows = min(PF_QT, S - q0);
   const int q_end = q0 + q_rows - 1;         // last (seq-local) q position
+  // chunked prefill: the chunk's q rows sit AFTER `off` cached keys; the
+  // k/v rows for this sequence span the whole context (prefix + chunk)
+  const int kv0 = kv_starts[seq];
+  const int KV = kv_starts[seq + 1] - kv0;
+  const int off = q_offs[seq];
 
   __shared__ ushort q_lds[PF_QT][PF_DPITCH];
   __shared__ ushort k_lds[PF_KT][PF_DPITCH];
@@ -96,8 +103,8 @@ void prefill_attention_kernel(
     for (int j = 0; j < 8; ++j) qfrag[dc][j] = (short)tmp.h[j];
   }
 
-  for (int k0 = 0; k0 <= q_end; k0 += PF_KT) {
-    const int kw = min(PF_KT, S - k0);
+  for (int k0 = 0; k0 <= off + q_end; k0 += PF_KT) {
+    const int kw = min(PF_KT, KV - k0);
     // ---- stage K and V tiles (coalesced; invalid rows zero) ----
     __syncthreads();
     for (int u = threadIdx.x; u < PF_KT * (PF_D / 8); u += blockDim.x) {
@@ -106,7 +113,7 @@ void prefill_attention_kernel(
       uint4 kv_ = make_uint4(0, 0, 0, 0);
       uint4 vv_ = make_uint4(0, 0, 0, 0);
       if (row < kw) {
-        const long long tok = (long long)(s0 + k0 + row);
+        const long long tok = (long long)(kv0 + k0 + row);
         kv_ = *reinterpret_cast<const uint4*>(
             k + tok * k_stride + (long long)hkv * PF_D + col);
         vv_ = *reinterpret_cast<const uint4*>(
@@ -153,7 +160,8 @@ void prefill_attention_kernel(
         const int row = wave * 16 + frag_kgrp * 4 + r;
         const int key = k0 + nt * 16 + frag_row;  // this lane's column
         float sv = s_acc[nt][r] * scale;
-        if (key > q0 + row || key >= S || row >= q_rows) sv = -INFINITY;
+        if (key > off + q0 + row || key >= KV || row >= q_rows)
+          sv = -INFINITY;
         p_val[nt][r] = sv;
       }
     }
@@ -254,6 +262,7 @@ void prefill_attention_kernel(
 void prefill_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                        torch::Tensor v, torch::Tensor tile_seq,
                        torch::Tensor tile_q0, torch::Tensor seq_starts,
+                       torch::Tensor kv_starts, torch::Tensor q_offs,
                        double scale) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(out.is_contiguous());
@@ -262,7 +271,9 @@ void prefill_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == PF_D);
   TORCH_CHECK(tile_seq.scalar_type() == torch::kInt32 &&
               tile_q0.scalar_type() == torch::kInt32 &&
-              seq_starts.scalar_type() == torch::kInt32);
+              seq_starts.scalar_type() == torch::kInt32 &&
+              kv_starts.scalar_type() == torch::kInt32 &&
+              q_offs.scalar_type() == torch::kInt32);
   const int Hq = q.size(1);
   const int Hkv = k.size(1);
   TORCH_CHECK(q.size(2) == PF_D, "head dim must be 128");
@@ -275,6 +286,7 @@ void prefill_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                      (const ushort*)q.data_ptr(), (const ushort*)k.data_ptr(),
                      (const ushort*)v.data_ptr(), tile_seq.data_ptr<int>(),
                      tile_q0.data_ptr<int>(), seq_starts.data_ptr<int>(),
+                     kv_starts.data_ptr<int>(), q_offs.data_ptr<int>(),
                      (float)scale, Hq, Hkv, (long long)q.stride(0),
                      (long long)k.stride(0), (long long)v.stride(0));
 }

@@ -27,6 +27,7 @@ class EngineConfig:
     kv_pages: int = 256            # pages per layer in the pool
     max_batch: int = 64
     max_model_len: int = 2048
+    max_prefill_tokens: int = 8192   # chunked-prefill budget per step
     seed: int = 0
     device: str = "cpu"
     tp_rank: int = 0
@@ -251,34 +252,67 @@ class Engine:
             s.finished = True
 
     def _step_prefill(self, seqs: list[Sequence]) -> dict[int, int]:
+        """Chunked prefill: up to cfg.max_prefill_tokens prompt tokens per
+        step.  A sequence whose prompt exceeds the remaining budget is
+        processed partially (its chunk attends the cached prefix via the
+        paged-cache gather) and produces no token until the last chunk."""
+        budget = self.cfg.max_prefill_tokens
+        sel: list[tuple[Sequence, int]] = []
+        total = 0
+        for s in seqs:
+            if total >= budget:
+                break
+            take = min(len(s.token_ids) - s.num_cached, budget - total)
+            sel.append((s, take))
+            total += take
         ids, pos, slots, starts = [], [], [], [0]
-        for s in seqs:
-            # chunked/continued prefill is not implemented: the in-chunk
-            # causal attention would ignore the cached prefix
-            assert s.num_cached == 0, "prefill must see the whole prompt"
-        for s in seqs:
-            self._ensure_pages(s, len(s.token_ids))
-            new = s.token_ids[s.num_cached:]
-            ids.extend(new)
-            pos.extend(range(s.num_cached, len(s.token_ids)))
-            for p in range(s.num_cached, len(s.token_ids)):
+        chunked = any(s.num_cached > 0 or
+                      take < len(s.token_ids) - s.num_cached
+                      for s, take in sel)
+        for s, take in sel:
+            end = s.num_cached + take
+            self._ensure_pages(s, end)
+            ids.extend(s.token_ids[s.num_cached:end])
+            pos.extend(range(s.num_cached, end))
+            for p in range(s.num_cached, end):
                 slots.append(s.pages[p // PAGE_SIZE] * PAGE_SIZE + p % PAGE_SIZE)
-            starts.append(starts[-1] + len(new))
+            starts.append(starts[-1] + take)
+        kv_row_idx = kv_starts = q_offsets = None
+        if chunked:
+            # gather index over the FULL context (prefix + chunk) per
+            # (token, head): flat cache row (page*Hkv + h)*page + off
+            hkv = self.kv_caches[0][0].size(1)
+            rows, kv_starts, q_offsets = [], [0], []
+            for s, take in sel:
+                end = s.num_cached + take
+                for t in range(end):
+                    base = s.pages[t // PAGE_SIZE] * hkv * PAGE_SIZE + \
+                        t % PAGE_SIZE
+                    for h in range(hkv):
+                        rows.append(base + h * PAGE_SIZE)
+                kv_starts.append(kv_starts[-1] + end)
+                q_offsets.append(s.num_cached)
+            kv_row_idx = torch.tensor(rows, device=self.device,
+                                      dtype=torch.int64)
         batch = PrefillBatch(
             input_ids=torch.tensor(ids, device=self.device, dtype=torch.long),
             positions=torch.tensor(pos, device=self.device, dtype=torch.int32),
             seq_starts=starts,
             slot_mapping=torch.tensor(slots, device=self.device,
-                                      dtype=torch.int64))
+                                      dtype=torch.int64),
+            kv_row_idx=kv_row_idx, kv_starts=kv_starts, q_offsets=q_offsets)
         hidden = self.model.forward_prefill(batch, self.kv_caches)
         logits = self.model.compute_logits(hidden)
-        next_tokens = self._sample(logits, seqs)
+        next_tokens = self._sample(logits, [s for s, _ in sel])
         out = {}
-        for i, s in enumerate(seqs):
-            s.num_cached = len(s.token_ids)
-            tok = int(next_tokens[i])
-            self._append_token(s, tok)
-            out[s.seq_id] = tok
+        for i, (s, take) in enumerate(sel):
+            s.num_cached += take
+            if s.num_cached == len(s.token_ids):
+                # prompt complete: this chunk's last hidden row is the
+                # real next-token position
+                tok = int(next_tokens[i])
+                self._append_token(s, tok)
+                out[s.seq_id] = tok
         return out
 
     def _step_decode(self, seqs: list[Sequence]) -> dict[int, int]:

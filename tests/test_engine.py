@@ -227,3 +227,52 @@ def test_sampling_stop_token_finishes_sequence():
     sid = e.add_request([3, 1, 4, 1, 5], SamplingParams(stop_token=first))
     e.step()
     assert e.sequences[sid].finished
+
+
+def test_chunked_prefill_matches_one_shot():
+    """A prompt prefilled in 4-token chunks must generate the same stream
+    as whole-prompt prefill (the chunk attends its cached prefix through
+    the paged-cache gather path)."""
+    from lws_amd.serving.engine import Engine, EngineConfig
+
+    prompt = list(range(3, 20))      # 17 tokens -> chunks of 4,4,4,4,1
+    full = Engine(EngineConfig(model="llama-tiny", device="cpu",
+                               kv_pages=64, seed=7))
+    full.load()
+    want = full.generate([prompt], max_new_tokens=5)[0]
+
+    chunked = Engine(EngineConfig(model="llama-tiny", device="cpu",
+                                  kv_pages=64, seed=7,
+                                  max_prefill_tokens=4))
+    chunked.load()
+    sid = chunked.add_request(prompt)
+    # 5 prefill chunks produce no token until the last; then 4 decodes
+    outs = []
+    for _ in range(5 + 4):
+        outs.append(chunked.step())
+    got = chunked.sequences[sid].token_ids[len(prompt):len(prompt) + 5]
+    assert got == want
+    # the first 4 chunk steps emitted nothing
+    assert all(o == {} for o in outs[:4])
+
+
+def test_chunked_prefill_mixed_batch():
+    """Budget splits a batch: a long prompt chunks across steps while a
+    short one completes and starts decoding."""
+    from lws_amd.serving.engine import Engine, EngineConfig
+
+    long_p, short_p = list(range(2, 14)), [5, 6, 7]
+    ref = Engine(EngineConfig(model="llama-tiny", device="cpu",
+                              kv_pages=64, seed=7))
+    ref.load()
+    want = ref.generate([long_p, short_p], max_new_tokens=3)
+
+    e = Engine(EngineConfig(model="llama-tiny", device="cpu", kv_pages=64,
+                            seed=7, max_prefill_tokens=5))
+    e.load()
+    sids = [e.add_request(long_p), e.add_request(short_p)]
+    for _ in range(12):
+        e.step()
+    got = [e.sequences[sids[0]].token_ids[len(long_p):len(long_p) + 3],
+           e.sequences[sids[1]].token_ids[len(short_p):len(short_p) + 3]]
+    assert got == want
