@@ -199,6 +199,53 @@ class Engine:
         if need > len(seq.pages):
             seq.pages.extend(self.allocator.alloc(need - len(seq.pages)))
 
+    # -- disaggregated prefill/decode KV handoff ------------------------
+    def _kv_row_index(self, seq: Sequence, upto: int) -> torch.Tensor:
+        """Flat cache-row index per (token, head): the KV caches are
+        [pages, Hkv, page, D], so row (page*Hkv + h)*page_size + off."""
+        hkv = self.kv_caches[0][0].size(1)
+        rows = []
+        for t in range(upto):
+            base = seq.pages[t // PAGE_SIZE] * hkv * PAGE_SIZE + t % PAGE_SIZE
+            for h in range(hkv):
+                rows.append(base + h * PAGE_SIZE)
+        return torch.tensor(rows, device=self.device, dtype=torch.int64)
+
+    def export_kv(self, seq_id: int) -> tuple[list[int], int, list]:
+        """Extract a sequence's cached KV for handoff to another engine
+        (the DS prefill -> decode data path).  Returns (token_ids,
+        num_cached, [L x (k [T, Hkv, D], v [T, Hkv, D])]).  Typically
+        called right after the prefill step, so token_ids ends with the
+        first generated token and num_cached covers the prompt."""
+        seq = self.sequences[seq_id]
+        T = seq.num_cached
+        idx = self._kv_row_index(seq, T)
+        d = self.model_cfg.head_dim
+        hkv = self.kv_caches[0][0].size(1)
+        layers = []
+        for k_cache, v_cache in self.kv_caches:
+            k = k_cache.reshape(-1, d).index_select(0, idx).view(T, hkv, d)
+            v = v_cache.reshape(-1, d).index_select(0, idx).view(T, hkv, d)
+            layers.append((k, v))
+        return list(seq.token_ids), T, layers
+
+    def import_kv(self, token_ids: list[int], num_cached: int, layers: list,
+                  sampling: Optional[SamplingParams] = None) -> int:
+        """Adopt a prefilled sequence: allocate pages, scatter the
+        transferred KV into this engine's pools, and resume at decode."""
+        sid = self.add_request(token_ids, sampling)
+        seq = self.sequences[sid]
+        self._ensure_pages(seq, len(token_ids))
+        seq.num_cached = num_cached
+        idx = self._kv_row_index(seq, num_cached)
+        d = self.model_cfg.head_dim
+        for (k_cache, v_cache), (k, v) in zip(self.kv_caches, layers):
+            k_cache.reshape(-1, d).index_copy_(
+                0, idx, k.to(self.device).reshape(-1, d))
+            v_cache.reshape(-1, d).index_copy_(
+                0, idx, v.to(self.device).reshape(-1, d))
+        return sid
+
     # -- scheduling: one engine step = prefill new seqs or decode all ---
     def step(self) -> dict[int, int]:
         """Returns {seq_id: next_token} for sequences that produced one.

@@ -1,0 +1,132 @@
+"""Disaggregated prefill->decode KV handoff (the DS roles data path):
+in-process handoff equivalence + 2-rank gloo send/recv."""
+import os
+import random
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_handoff_local_equivalence():
+    """prefill on engine A -> handoff -> decode on engine B must produce
+    the same stream as a single engine doing both."""
+    from lws_amd.serving.engine import Engine, EngineConfig
+    from lws_amd.serving.handoff import handoff_local
+
+    prompt = list(range(3, 17))
+    single = Engine(EngineConfig(model="llama-tiny", device="cpu",
+                                 kv_pages=64, seed=7))
+    single.load()
+    want = single.generate([prompt], max_new_tokens=5)[0]
+
+    prefill = Engine(EngineConfig(model="llama-tiny", device="cpu",
+                                  kv_pages=64, seed=7))
+    prefill.load()
+    decode = Engine(EngineConfig(model="llama-tiny", device="cpu",
+                                 kv_pages=64, seed=7))
+    decode.load()
+    sid = prefill.add_request(prompt)
+    first = prefill.step()[sid]           # prefill emits the first token
+    did = handoff_local(prefill, decode, sid)
+    assert sid not in prefill.sequences   # freed on the prefill side
+    for _ in range(4):
+        decode.step()
+    got = [first] + decode.sequences[did].token_ids[len(prompt) + 1:]
+    assert got == want
+
+
+def test_handoff_pages_released():
+    from lws_amd.serving.engine import Engine, EngineConfig
+    from lws_amd.serving.handoff import handoff_local
+
+    prefill = Engine(EngineConfig(model="llama-tiny", device="cpu",
+                                  kv_pages=8, seed=7))
+    prefill.load()
+    decode = Engine(EngineConfig(model="llama-tiny", device="cpu",
+                                 kv_pages=64, seed=7))
+    decode.load()
+    free0 = len(prefill.allocator.free)
+    for i in range(6):                     # would exhaust 8 pages w/o free
+        sid = prefill.add_request(list(range(2, 2 + 20)))
+        prefill.step()
+        handoff_local(prefill, decode, sid)
+    assert len(prefill.allocator.free) == free0
+
+
+WORKER = r"""
+import os, sys
+sys.path.insert(0, os.environ["LWS_REPO"])
+import torch.distributed as dist
+from lws_amd.serving.engine import Engine, EngineConfig
+from lws_amd.serving.handoff import recv_kv, send_kv
+
+dist.init_process_group("gloo")
+rank = dist.get_rank()
+prompt = list(range(3, 17))
+eng = Engine(EngineConfig(model="llama-tiny", device="cpu", kv_pages=64,
+                          seed=7))
+eng.load()
+if rank == 0:
+    sid = eng.add_request(prompt)
+    first = eng.step()[sid]
+    send_kv(eng, sid, dst=1)
+    dist.barrier()
+else:
+    want = eng.generate([prompt], max_new_tokens=5)[0]  # local reference
+    did = recv_kv(eng, src=0)
+    seq = eng.sequences[did]
+    assert seq.num_cached == len(prompt), seq.num_cached
+    for _ in range(4):
+        eng.step()
+    got = seq.token_ids[len(prompt):len(prompt) + 5]
+    assert got == want, (got, want)
+    print("HANDOFF_OK", flush=True)
+    dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_handoff_2rank_gloo(tmp_path):
+    script = tmp_path / "handoff_worker.py"
+    script.write_text(WORKER)
+    env = dict(os.environ, LWS_REPO=REPO)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(random.randint(21000, 59000)), str(script)],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert "HANDOFF_OK" in out.stdout
+
+
+@pytest.mark.gpu
+def test_handoff_local_equivalence_gpu():
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from lws_amd.serving.engine import Engine, EngineConfig
+    from lws_amd.serving.handoff import handoff_local
+
+    prompt = list(range(3, 40))
+    single = Engine(EngineConfig(model="llama-tiny", device="cuda",
+                                 kv_pages=64, seed=7))
+    single.load()
+    want = single.generate([prompt], max_new_tokens=5)[0]
+    single.unload()
+
+    prefill = Engine(EngineConfig(model="llama-tiny", device="cuda",
+                                  kv_pages=64, seed=7))
+    prefill.load()
+    decode = Engine(EngineConfig(model="llama-tiny", device="cuda",
+                                 kv_pages=64, seed=7))
+    decode.load()
+    sid = prefill.add_request(prompt)
+    first = prefill.step()[sid]
+    did = handoff_local(prefill, decode, sid)
+    for _ in range(4):
+        decode.step()
+    got = [first] + decode.sequences[did].token_ids[len(prompt) + 1:]
+    assert got == want
