@@ -102,7 +102,8 @@ __device__ __forceinline__ void sg_wait_vm() {
 
 template <int MTILES, bool XLDS>
 __global__ __launch_bounds__(256)
-void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
+void skinny_gemm_kernel(ushort* __restrict__ out,       // [M, N] bf16
+                        float* __restrict__ out_ws,      // [splits, M, N] fp32
                         const ushort* __restrict__ x,    // [M, K]
                         const ushort* __restrict__ w,    // [N, K]
                         int M, int N, int K, int k_slice) {
@@ -225,6 +226,27 @@ void skinny_gemm_kernel(float* __restrict__ out_ws,      // [splits, M, N] fp32
 
   // C/D layout: lane l reg r -> D[row=(l/16)*4+r][col=l%16].
   const int n = n0 + wave * SG_NTILE + frag_row;
+
+  if (gridDim.y == 1) {
+    // unsplit: write bf16 straight out — no workspace, no reduction
+    if (n < N) {
+#pragma unroll
+      for (int t = 0; t < MTILES; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = t * 16 + frag_kgrp * 4 + r;
+          if (m < M) out[(long long)m * N + n] = f32_to_bf16(acc[t][r]);
+        }
+      }
+    }
+    return;
+  }
+
+  // split path: fp32 partial plane + separate finalize kernel.  (A fused
+  // last-block reduction was tried and measured 3x SLOWER: the release/
+  // acquire __threadfence pair per block writes back + invalidates the
+  // per-XCD L2, and the tail is one latency-bound workgroup per n-range
+  // — gpurun_out/sg_fused.log.)
   float* plane = out_ws + (long long)blockIdx.y * M * N;
   if (n < N) {
 #pragma unroll
@@ -290,7 +312,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const bool xlds = env_int("LWS_SG_XLDS", 1) != 0;
 #define SG_LAUNCH(MT, XL)                                                    \
   hipLaunchKernelGGL((skinny_gemm_kernel<MT, XL>), grid, dim3(256), 0,        \
-                     stream, ws.data_ptr<float>(),                            \
+                     stream, (ushort*)out.data_ptr(), ws.data_ptr<float>(),   \
                      (const ushort*)x.data_ptr(),                             \
                      (const ushort*)w.data_ptr(), M, N, K, k_slice)
   if (M <= 16) {
@@ -299,9 +321,11 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
     if (xlds) SG_LAUNCH(2, true); else SG_LAUNCH(2, false);
   }
 #undef SG_LAUNCH
-  long long total = (long long)M * N;
-  long long blocks = min((total + 255) / 256, (long long)2048);
-  hipLaunchKernelGGL(skinny_gemm_finalize_kernel, dim3((int)blocks), dim3(256),
-                     0, stream, (ushort*)out.data_ptr(), ws.data_ptr<float>(),
-                     total, grid_y);
+  if (grid_y > 1) {
+    long long total = (long long)M * N;
+    long long blocks = min((total + 255) / 256, (long long)2048);
+    hipLaunchKernelGGL(skinny_gemm_finalize_kernel, dim3((int)blocks),
+                       dim3(256), 0, stream, (ushort*)out.data_ptr(),
+                       ws.data_ptr<float>(), total, grid_y);
+  }
 }
