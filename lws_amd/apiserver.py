@@ -59,12 +59,15 @@ def build_app(store: Store, manager=None):
     def metrics():
         from fastapi.responses import PlainTextResponse
 
+        from .cluster.metrics import GLOBAL as reconcile_metrics
+
         lines = ["# TYPE lws_amd_objects gauge"]
         counts: dict[str, int] = {}
         for kind, ns, name in store.snapshot_keys():
             counts[kind] = counts.get(kind, 0) + 1
         for kind, n in sorted(counts.items()):
             lines.append(f'lws_amd_objects{{kind="{kind}"}} {n}')
+        lines.extend(reconcile_metrics.render())
         return PlainTextResponse("\n".join(lines) + "\n")
 
     @app.post("/apis/{resource}/namespaces/{ns}")

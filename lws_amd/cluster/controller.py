@@ -17,6 +17,7 @@ import time
 import traceback
 from typing import Callable, Optional
 
+from .metrics import GLOBAL as metrics
 from .store import ConflictError, Store
 
 log = logging.getLogger("lws_amd")
@@ -124,18 +125,25 @@ class Controller:
                     return
                 continue
             ns, name = key
+            t0 = time.perf_counter()
             try:
                 result = self.reconcile(ns, name)
                 if isinstance(result, (int, float)) and result > 0:
                     self.queue.add_after(key, float(result))
+                    outcome = "requeue_after"
+                else:
+                    outcome = "success"
             except ConflictError:
                 self.queue.add_after(key, 0.01)
+                outcome = "conflict_requeue"
             except Exception:  # noqa: BLE001 — reconcile errors retry w/ backoff
                 log.error("reconcile %s %s/%s failed:\n%s", self.name, ns, name,
                           traceback.format_exc())
                 self.queue.add_after(key, 0.2)
+                outcome = "error"
             finally:
                 self.queue.done(key)
+            metrics.observe(self.name, time.perf_counter() - t0, outcome)
 
     def stop(self) -> None:
         self._stop.set()

@@ -151,3 +151,30 @@ def test_ds_over_http(manager_proc):
     ds_client.delete("http-ds")
     wait_for(lambda: not cs.leader_worker_sets("default").list(),
              desc="children GC'd", timeout=30)
+
+
+def test_metrics_endpoint_reconcile_families(manager_proc):
+    """/metrics serves the controller-runtime-parity reconcile families
+    (reconcile_total, errors_total, time_seconds histogram) plus object
+    gauges after activity."""
+    import httpx
+
+    from lws_amd.client.clientset import Clientset
+    from tests.conftest import make_lws, wait_for
+
+    _, base = manager_proc
+    cs = Clientset.for_server(base)
+    cs.leader_worker_sets().create(make_lws(name="metrics-lws", replicas=1,
+                                            size=2))
+
+    def ready():
+        cur = cs.leader_worker_sets().get("metrics-lws")
+        conds = {c.type: c.status for c in cur.status.conditions}
+        return cur if conds.get("Available") == "True" else None
+    wait_for(ready, desc="metrics-lws Available", timeout=60)
+
+    body = httpx.get(base + "/metrics", timeout=10).text
+    assert 'lws_amd_objects{kind="LeaderWorkerSet"} 1' in body
+    assert 'lws_amd_reconcile_total{controller="leaderworkerset"' in body
+    assert "lws_amd_reconcile_time_seconds_bucket" in body
+    assert "lws_amd_reconcile_time_seconds_count" in body
