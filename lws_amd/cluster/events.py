@@ -37,10 +37,38 @@ class Event:
 
 
 class EventRecorder:
-    """events.EventRecorder equivalent backed by the object store."""
+    """events.EventRecorder equivalent backed by the object store.
 
-    def __init__(self, store):
+    Events expire after ``ttl_seconds`` (kube's default event TTL is 1 h)
+    and the total is bounded by ``max_events`` — pruning runs inline on
+    record, oldest-last-timestamp first, so an event storm cannot grow
+    the store without bound."""
+
+    def __init__(self, store, ttl_seconds: float = 3600.0,
+                 max_events: int = 1000):
         self.store = store
+        self.ttl = ttl_seconds
+        self.max_events = max_events
+        self._last_prune = 0.0
+
+    def _prune(self, now: float) -> None:
+        if now - self._last_prune < 5.0:
+            return
+        self._last_prune = now
+        from .store import ApiError
+
+        events = self.store.list("Event")
+        expired = [e for e in events if now - e.last_timestamp > self.ttl]
+        keep = [e for e in events if now - e.last_timestamp <= self.ttl]
+        if len(keep) > self.max_events:
+            keep.sort(key=lambda e: e.last_timestamp)
+            expired.extend(keep[:len(keep) - self.max_events])
+        for e in expired:
+            try:
+                self.store.delete("Event", e.metadata.namespace,
+                                  e.metadata.name)
+            except ApiError:
+                pass
 
     def eventf(self, obj, event_type: str, reason: str, message: str) -> None:
         from .store import AlreadyExistsError, ConflictError, NotFoundError
@@ -51,6 +79,7 @@ class EventRecorder:
         # aggregate by (object, reason) like kube's event correlation
         name = f"{ref.name}.{reason.lower()}"
         now = time.time()
+        self._prune(now)
         for _ in range(5):
             existing = self.store.try_get("Event", ref.namespace, name)
             if existing is None:

@@ -138,3 +138,30 @@ def test_delete_during_reconcile_leaves_nothing(cluster):
                 if k[0] != "Event"]
         return (not keys) or None
     wait_for(clean, desc="no leaked objects after churn", timeout=40)
+
+
+def test_event_ttl_and_cap_pruning():
+    """Events expire by TTL and the total is capped (kube event TTL
+    analogue) so an event storm can't grow the store unboundedly."""
+    import time as _time
+
+    from lws_amd.cluster.events import EventRecorder
+    from lws_amd.cluster.store import Store
+    from tests.conftest import make_lws
+
+    store = Store()
+    rec = EventRecorder(store, ttl_seconds=0.2, max_events=10)
+    lws = make_lws(name="evt")
+    lws.metadata.namespace = "default"
+    lws.metadata.uid = "u1"
+    for i in range(30):
+        o = make_lws(name=f"evt-{i}")
+        o.metadata.namespace = "default"
+        rec.eventf(o, "Normal", "Thing", f"m{i}")
+    # cap enforcement happens on the next prune window
+    _time.sleep(0.25)
+    rec._last_prune = 0.0
+    rec.eventf(lws, "Normal", "Tick", "t")
+    evs = store.list("Event")
+    # all 30 old events expired via TTL; only the fresh one remains
+    assert len(evs) == 1 and evs[0].reason == "Tick"
