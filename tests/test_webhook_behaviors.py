@@ -125,8 +125,16 @@ def test_recreate_group_after_start_gate():
         from lws_amd.cluster.store import ConflictError
 
         def bump_restart(name, phase="Running"):
+            # the pod can be mid-recreate when Available flips; wait for it
+            wait_for(lambda: c.store.try_get("Pod", "default", name),
+                     desc=f"{name} exists", timeout=20)
+            from lws_amd.cluster.store import NotFoundError
             for _ in range(50):
-                cur = c.store.get("Pod", "default", name)
+                try:
+                    cur = c.store.get("Pod", "default", name)
+                except NotFoundError:
+                    time.sleep(0.05)
+                    continue
                 cur.status.phase = phase
                 if cur.status.container_statuses:
                     cur.status.container_statuses[0].restart_count += 1
