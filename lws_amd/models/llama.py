@@ -178,6 +178,20 @@ class LlamaLayer:
                 ShardedLinear(h, cfg.intermediate_size, 1, tp_rank, tp_world,
                               device, dtype)
                 for _ in range(cfg.num_experts)]
+            # stack each expert family into ONE [E, N, K] tensor and make
+            # the per-expert weights views into it: the grouped skinny
+            # GEMM then streams every expert's weights in a single launch
+            # (decode reads ALL active experts anyway — E serial skinny/
+            # hipBLASLt calls were ~2.5 TB/s each, chip underfilled)
+            def stack(lins):
+                w0 = lins[0].weight
+                stacked = torch.empty(cfg.num_experts, *w0.shape,
+                                      device=device, dtype=dtype)
+                for i, lin in enumerate(lins):
+                    lin.weight = stacked[i]
+                return stacked
+            self.experts_gate_up_w = stack(self.experts_gate_up)
+            self.experts_down_w = stack(self.experts_down)
             self.gate_up = None
             self.down = None
         else:
@@ -431,6 +445,32 @@ class LlamaForCausalLM:
                                      cfg.num_experts_per_tok, dim=-1)
         weights = weights / weights.sum(dim=-1, keepdim=True)
         out = torch.zeros_like(h)
+        if self._ops.is_gpu and h.size(0) <= 32:
+            # decode: grouped expert GEMMs — every expert's tokens padded
+            # into [E, 32, K] and ONE kernel launch per projection streams
+            # all expert weights concurrently (vs E serial chip-underfilled
+            # launches; see profiles/r01_skinny_dispatch.md MoE section)
+            E = cfg.num_experts
+            Mp = h.size(0)
+            x_pad = torch.zeros(E, Mp, h.size(1), dtype=h.dtype,
+                                device=h.device)
+            token_lists = []
+            for e in range(E):
+                t_idx, s_idx = (chosen == e).nonzero(as_tuple=True)
+                token_lists.append((t_idx, s_idx))
+                if t_idx.numel():
+                    x_pad[e, :t_idx.numel()] = h[t_idx]
+            from .. import ops
+            gu = ops.skinny_gemm(x_pad, layer.experts_gate_up_w)
+            act = self._silu_mul(gu.view(E * Mp, -1)).view(E, Mp, -1)
+            dn = ops.skinny_gemm(act.contiguous(), layer.experts_down_w)
+            for e in range(E):
+                t_idx, s_idx = token_lists[e]
+                if not t_idx.numel():
+                    continue
+                wgt = weights[t_idx, s_idx].unsqueeze(-1).to(h.dtype)
+                out.index_add_(0, t_idx, dn[e, :t_idx.numel()] * wgt)
+            return out
         for e in range(cfg.num_experts):
             mask = (chosen == e)
             token_idx, slot_idx = mask.nonzero(as_tuple=True)

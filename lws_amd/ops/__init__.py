@@ -162,24 +162,34 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor,
                 out: Optional[torch.Tensor] = None) -> torch.Tensor:
     """out[M,N] = x[M,K] @ w[N,K]^T for M <= 32 — split-K MFMA kernel that
     fills all 256 CUs where hipBLASLt's heuristic tiles underfill the chip
-    on decode-shape GEMMs (see profiles/)."""
+    on decode-shape GEMMs (see profiles/).
+
+    Grouped (MoE) form: x [E,M,K] @ w [E,N,K]^T -> out [E,M,N] — one
+    launch covers every expert, so all expert weights stream at full chip
+    parallelism instead of E serial skinny launches."""
     lib = require_native()
-    M, K = x.shape
-    N = w.size(0)
+    grouped = x.dim() == 3
+    E = x.size(0) if grouped else 1
+    M, K = x.shape[-2], x.shape[-1]
+    N = w.size(-2)
     if out is None:
-        out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+        shape = (E, M, N) if grouped else (M, N)
+        out = torch.empty(shape, dtype=x.dtype, device=x.device)
     import os
     target = int(os.environ.get("LWS_SG_TARGET", "256"))  # keep = C++ default
     n_blocks = (N + 63) // 64
     split = min(max(1, target // max(1, n_blocks)), max(1, K // 128))
     k_slice = (K // split + 127) // 128 * 128
     grid_y = (K + k_slice - 1) // k_slice
-    key = (grid_y, M, N, x.device.index)
+    if grouped and E * n_blocks >= 256:
+        grid_y = 1        # mirror the C++ grouped no-split policy
+    key = (grid_y, E, M, N, x.device.index)
     ws = _SKINNY_WS.get(key)
     if ws is None:
         # grid_y == 1 writes bf16 directly and never touches ws, but the
         # C++ side still size-checks it — keep the uniform allocation
-        ws = torch.empty(grid_y, M, N, dtype=torch.float32, device=x.device)
+        ws = torch.empty(grid_y, E * M, N, dtype=torch.float32,
+                         device=x.device)
         _SKINNY_WS[key] = ws
     lib.skinny_gemm(out, x, w, ws)
     return out

@@ -112,6 +112,12 @@ void skinny_gemm_kernel(ushort* __restrict__ out,       // [M, N] bf16
   const int n0 = blockIdx.x * SG_ROWS;
   const int kbegin = blockIdx.y * k_slice;
   const int kend = min(kbegin + k_slice, K);
+  // grouped (MoE) launch: blockIdx.z selects the expert; x/w/out are
+  // [E, M, K] / [E, N, K] / [E, M, N] and E=1 for the plain call
+  const int e = blockIdx.z;
+  x += (long long)e * M * K;
+  w += (long long)e * N * K;
+  out += (long long)e * M * N;
 
   const int frag_row = lane % 16;        // m (A) / n (B)
   const int frag_kgrp = lane / 16;       // which 8-wide k group
@@ -247,7 +253,8 @@ void skinny_gemm_kernel(ushort* __restrict__ out,       // [M, N] bf16
   // acquire __threadfence pair per block writes back + invalidates the
   // per-XCD L2, and the tail is one latency-bound workgroup per n-range
   // — gpurun_out/sg_fused.log.)
-  float* plane = out_ws + (long long)blockIdx.y * M * N;
+  float* plane = out_ws +
+      ((long long)blockIdx.y * gridDim.z + e) * M * N;
   if (n < N) {
 #pragma unroll
     for (int t = 0; t < MTILES; ++t) {
@@ -286,10 +293,18 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(ws.scalar_type() == torch::kFloat32 && ws.is_contiguous());
-  const int M = x.size(0);
-  const int K = x.size(1);
-  const int N = w.size(0);
-  TORCH_CHECK(w.size(1) == K && out.size(0) == M && out.size(1) == N);
+  // grouped (MoE) form: x [E, M, K], w [E, N, K], out [E, M, N]
+  const bool grouped = x.dim() == 3;
+  const int E = grouped ? x.size(0) : 1;
+  if (grouped)
+    TORCH_CHECK(w.dim() == 3 && out.dim() == 3 && w.size(0) == E &&
+                out.size(0) == E);
+  const int M = x.size(grouped ? 1 : 0);
+  const int K = x.size(grouped ? 2 : 1);
+  const int N = w.size(grouped ? 1 : 0);
+  TORCH_CHECK(w.size(grouped ? 2 : 1) == K &&
+              out.size(grouped ? 1 : 0) == M &&
+              out.size(grouped ? 2 : 1) == N);
   TORCH_CHECK(M <= 32, "skinny_gemm: M must be <= 32");
   TORCH_CHECK(K % SG_KSUB == 0 && N % 16 == 0,
               "skinny_gemm: K must be a multiple of 128");
@@ -304,11 +319,17 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   if (split > max_split) split = max_split;
   if (split < 1) split = 1;
   int k_slice = (K / split + SG_KSUB - 1) / SG_KSUB * SG_KSUB;
-  const int grid_y = (K + k_slice - 1) / k_slice;
-  TORCH_CHECK(ws.numel() >= (long long)grid_y * M * N,
+  int grid_y = (K + k_slice - 1) / k_slice;
+  if (grouped && E * n_blocks >= 256) {
+    // the expert dimension already fills the chip: no split, direct
+    // bf16 writes, full-depth K pipeline per block
+    grid_y = 1;
+    k_slice = (K + SG_KSUB - 1) / SG_KSUB * SG_KSUB;
+  }
+  TORCH_CHECK(ws.numel() >= (long long)grid_y * E * M * N,
               "skinny_gemm workspace too small");
 
-  dim3 grid(n_blocks, grid_y);
+  dim3 grid(n_blocks, grid_y, E);
   const bool xlds = env_int("LWS_SG_XLDS", 1) != 0;
 #define SG_LAUNCH(MT, XL)                                                    \
   hipLaunchKernelGGL((skinny_gemm_kernel<MT, XL>), grid, dim3(256), 0,        \
@@ -322,7 +343,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   }
 #undef SG_LAUNCH
   if (grid_y > 1) {
-    long long total = (long long)M * N;
+    long long total = (long long)E * M * N;
     long long blocks = min((total + 255) / 256, (long long)2048);
     hipLaunchKernelGGL(skinny_gemm_finalize_kernel, dim3((int)blocks),
                        dim3(256), 0, stream, (ushort*)out.data_ptr(),

@@ -316,3 +316,23 @@ def test_chunked_prefill_engine_gpu():
     got = chunked.sequences[sid].token_ids[len(prompt):len(prompt) + 5]
     chunked.unload()
     assert got == want
+
+
+@gpu
+@requires_gpu
+def test_skinny_gemm_grouped():
+    """Grouped (MoE) form: one launch over E experts == per-expert fp32
+    reference matmuls."""
+    import lws_amd.ops as ops
+
+    torch.manual_seed(0)
+    for (E, M, N, K) in [(8, 32, 1792, 4096), (8, 8, 4096, 1792),
+                         (4, 16, 28672, 4096), (8, 32, 4096, 14336)]:
+        x = torch.randn(E, M, K, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(E, N, K, dtype=torch.bfloat16, device="cuda")
+        out = ops.skinny_gemm(x, w)
+        assert out.shape == (E, M, N)
+        for e in range(E):
+            ref = x[e].float() @ w[e].float().t()
+            err = (out[e].float() - ref).abs().max().item()
+            assert err < 0.15 * K ** 0.5, (E, M, N, K, e, err)
