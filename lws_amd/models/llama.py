@@ -447,30 +447,31 @@ class LlamaForCausalLM:
         out = torch.zeros_like(h)
         if self._ops.is_gpu and h.size(0) <= 32:
             # decode: grouped expert GEMMs — every expert's tokens padded
-            # into [E, 32, K] and ONE kernel launch per projection streams
+            # into [E, T, K] and ONE kernel launch per projection streams
             # all expert weights concurrently (vs E serial chip-underfilled
-            # launches; see profiles/r01_skinny_dispatch.md MoE section)
+            # launches).  The routing scatter/gather is fully tensorized —
+            # no nonzero()/item() host syncs — so the whole MoE step stays
+            # hipGraph-capturable.
             E = cfg.num_experts
-            Mp = h.size(0)
-            x_pad = torch.zeros(E, Mp, h.size(1), dtype=h.dtype,
+            T = h.size(0)
+            tk = cfg.num_experts_per_tok
+            flat = chosen.reshape(-1)                       # [T*tk]
+            oh = torch.nn.functional.one_hot(flat, E)       # [T*tk, E]
+            ranks = oh.cumsum(0) - oh                       # earlier same-e
+            rank_sel = ranks.gather(1, flat.unsqueeze(1)).squeeze(1)
+            dest = flat * T + rank_sel                      # row in [E*T]
+            tok = torch.arange(T, device=h.device).repeat_interleave(tk)
+            x_pad = torch.zeros(E * T, h.size(1), dtype=h.dtype,
                                 device=h.device)
-            token_lists = []
-            for e in range(E):
-                t_idx, s_idx = (chosen == e).nonzero(as_tuple=True)
-                token_lists.append((t_idx, s_idx))
-                if t_idx.numel():
-                    x_pad[e, :t_idx.numel()] = h[t_idx]
+            x_pad.index_copy_(0, dest, h.index_select(0, tok))
             from .. import ops
-            gu = ops.skinny_gemm(x_pad, layer.experts_gate_up_w)
-            act = self._silu_mul(gu.view(E * Mp, -1)).view(E, Mp, -1)
-            dn = ops.skinny_gemm(act.contiguous(), layer.experts_down_w)
-            for e in range(E):
-                t_idx, s_idx = token_lists[e]
-                if not t_idx.numel():
-                    continue
-                wgt = weights[t_idx, s_idx].unsqueeze(-1).to(h.dtype)
-                out.index_add_(0, t_idx, dn[e, :t_idx.numel()] * wgt)
-            return out
+            gu = ops.skinny_gemm(x_pad.view(E, T, -1),
+                                 layer.experts_gate_up_w)
+            act = self._silu_mul(gu.view(E * T, -1))
+            dn = ops.skinny_gemm(act.view(E, T, -1), layer.experts_down_w)
+            y = dn.view(E * T, -1).index_select(0, dest)    # [T*tk, H]
+            wgt = weights.reshape(-1, 1).to(h.dtype)
+            return (y * wgt).view(T, tk, -1).sum(1)
         for e in range(cfg.num_experts):
             mask = (chosen == e)
             token_idx, slot_idx = mask.nonzero(as_tuple=True)

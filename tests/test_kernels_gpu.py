@@ -336,3 +336,40 @@ def test_skinny_gemm_grouped():
             ref = x[e].float() @ w[e].float().t()
             err = (out[e].float() - ref).abs().max().item()
             assert err < 0.15 * K ** 0.5, (E, M, N, K, e, err)
+
+
+@gpu
+@requires_gpu
+def test_moe_grouped_mlp_matches_reference():
+    """The tensorized grouped MoE decode path == per-expert fp32 loop."""
+    from lws_amd.models.llama import MODEL_PRESETS, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(MODEL_PRESETS["mixtral-tiny"](), device="cuda")
+    model.materialize(seed=3)
+    layer = model.layers[0]
+    cfg = model.cfg
+    h = torch.randn(17, cfg.hidden_size, dtype=torch.bfloat16,
+                    device="cuda")
+    got = model._mlp(layer, h).float()
+
+    # fp32 reference: same router math, per-expert loop
+    logits = h.float() @ layer.router.float().t()
+    weights, chosen = torch.topk(torch.softmax(logits, -1),
+                                 cfg.num_experts_per_tok, -1)
+    weights = weights / weights.sum(-1, keepdim=True)
+    want = torch.zeros(h.size(0), cfg.hidden_size, device="cuda")
+    for e in range(cfg.num_experts):
+        t_idx, s_idx = (chosen == e).nonzero(as_tuple=True)
+        if not t_idx.numel():
+            continue
+        xe = h[t_idx].float()
+        guw = layer.experts_gate_up_w[e].float()
+        dnw = layer.experts_down_w[e].float()
+        gu2 = xe @ guw.t()
+        g, u = gu2.chunk(2, -1)
+        ye = (torch.nn.functional.silu(g) * u) @ dnw.t()
+        want.index_add_(0, t_idx,
+                        ye * weights[t_idx, s_idx].unsqueeze(-1))
+    err = (got - want).abs().max().item()
+    assert err < 0.3, err
