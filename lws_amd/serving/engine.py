@@ -476,6 +476,7 @@ class _CapturedDecode:
         torch.argmax(logits, dim=-1, out=self.tokens)
 
     def _capture(self) -> None:
+        import gc
         import logging
 
         eng = self.engine
@@ -484,9 +485,17 @@ class _CapturedDecode:
             for _ in range(2):
                 self._forward()
             torch.cuda.synchronize(eng.device)
-            self.graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(self.graph):
-                self._forward()
+            # quiesce the Python GC for the capture: a GC cycle firing
+            # mid-capture runs tensor __del__ -> hipFree on the capturing
+            # stream, which ABORTS the process (not a catchable error)
+            gc.collect()
+            gc.disable()
+            try:
+                self.graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(self.graph):
+                    self._forward()
+            finally:
+                gc.enable()
         except Exception as e:  # noqa: BLE001 — capture-unsupported path
             logging.getLogger("lws_amd").warning(
                 "hipGraph capture failed (%s); decode runs eager", e)
