@@ -45,6 +45,12 @@ class LwsCluster:
                  enable_ds: bool = True) -> None:
         self.manager = Manager()
         self.store: Store = self.manager.store
+        # index the hot selector keys: every pod/STS/service lookup the
+        # reconcilers make filters on the set-name label — at 10^3+ groups
+        # a full-kind scan per reconcile dominates
+        from ..api import leaderworkerset as lwsapi
+        self.store.add_label_index(lwsapi.SET_NAME_LABEL_KEY)
+        self.store.add_label_index(lwsapi.GROUP_INDEX_LABEL_KEY)
         self.nodes = nodes if nodes is not None else make_nodes(1)
         self._node_by_name = {n.metadata.name: n for n in self.nodes}
 

@@ -48,7 +48,10 @@ class Scheduler:
                    if not p.node_name and p.metadata.deletion_timestamp is None]
         if not pending:
             return None
-        scheduled = [p for p in self.store.list("Pod", namespace) if p.node_name]
+        # read-only lister refs: `scheduled` is only inspected for
+        # affinity/capacity accounting, never mutated
+        scheduled = [p for p in self.store.list("Pod", namespace, copy=False)
+                     if p.node_name]
 
         # partition into gang groups and singletons
         gangs: dict[str, list[Pod]] = {}

@@ -60,6 +60,7 @@ def template_hash(template) -> str:
 class StatefulSetController:
     def __init__(self, manager: Manager) -> None:
         self.store: Store = manager.store
+        self.store.add_label_index(STS_OWNER_LABEL)
         self.ctrl = Controller("statefulset", self.reconcile)
         manager.add_controller(self.ctrl)
         manager.watch("StatefulSet", self.ctrl)
@@ -140,9 +141,12 @@ class StatefulSetController:
     # ------------------------------------------------------------------
     def _owned_pods(self, sts: StatefulSet) -> dict[int, Pod]:
         out: dict[int, Pod] = {}
-        for pod in self.store.list("Pod", sts.metadata.namespace):
-            if pod.metadata.labels.get(STS_OWNER_LABEL) != sts.metadata.name:
-                continue
+        # indexed lookup + shared refs (read-only: every pod mutation in
+        # this controller goes through store verbs on fresh objects)
+        for pod in self.store.list(
+                "Pod", sts.metadata.namespace,
+                label_selector={STS_OWNER_LABEL: sts.metadata.name},
+                copy=False):
             parent, ord_ = parse_parent_and_ordinal(pod.metadata.name)
             if parent != sts.metadata.name or ord_ < 0:
                 continue

@@ -72,6 +72,10 @@ class Store:
         self._objects: dict[tuple[str, str, str], Any] = {}
         # secondary index: kind -> {key -> obj} (shared object refs)
         self._by_kind: dict[str, dict[tuple[str, str, str], Any]] = {}
+        # label index: label_key -> {(kind, ns, value) -> set of obj keys}
+        # (client-go field-index analogue; register hot selector keys with
+        # add_label_index — list() then skips full-kind scans)
+        self._label_index: dict[str, dict[tuple, set]] = {}
         self._rv = 0
         self._uid = 0
         # kind -> list of fn(event_type, obj)
@@ -88,6 +92,26 @@ class Store:
 
     def add_validator(self, kind: str, fn: Callable[[Any, Optional[Any]], None]) -> None:
         self._validators.setdefault(kind, []).append(fn)
+
+    def add_label_index(self, label_key: str) -> None:
+        """Index objects by a label key (register at assembly time)."""
+        self._label_index.setdefault(label_key, {})
+
+    def _index_add(self, key, obj) -> None:
+        labels = obj.metadata.labels or {}
+        for lk, idx in self._label_index.items():
+            if lk in labels:
+                idx.setdefault((key[0], key[1], labels[lk]), set()).add(key)
+
+    def _index_remove(self, key, obj) -> None:
+        if obj is None:
+            return
+        labels = obj.metadata.labels or {}
+        for lk, idx in self._label_index.items():
+            if lk in labels:
+                bucket = idx.get((key[0], key[1], labels[lk]))
+                if bucket is not None:
+                    bucket.discard(key)
 
     def add_handler(self, kind: Optional[str], fn: Callable[[str, Any], None]) -> None:
         """Register a watch handler; kind=None receives all events."""
@@ -129,6 +153,7 @@ class Store:
             obj.metadata.creation_timestamp = time.time()
             obj.metadata.deletion_timestamp = None
             self._objects[key] = obj
+            self._index_add(key, obj)
             self._by_kind.setdefault(key[0], {})[key] = obj
             events.append((ADDED, obj))
         self._dispatch(events)
@@ -149,10 +174,29 @@ class Store:
 
     def list(self, kind: str, namespace: Optional[str] = None,
              label_selector: Optional[dict[str, str]] = None,
-             filter_fn: Optional[Callable[[Any], bool]] = None) -> list[Any]:
+             filter_fn: Optional[Callable[[Any], bool]] = None,
+             copy: bool = True) -> list[Any]:
+        """copy=False returns SHARED object refs (client-go lister
+        semantics): the caller must treat them as read-only — mutations go
+        through update()/update_status() on fresh copies."""
         out = []
         with self._lock:
-            for (k, ns, _), obj in self._by_kind.get(kind, {}).items():
+            candidates = None
+            if label_selector and namespace is not None:
+                best = None
+                for lk, lv in label_selector.items():
+                    idx = self._label_index.get(lk)
+                    if idx is not None:
+                        bucket = idx.get((kind, namespace, lv), set())
+                        if best is None or len(bucket) < len(best):
+                            best = bucket
+                if best is not None:
+                    candidates = [self._objects[k2] for k2 in list(best)
+                                  if k2 in self._objects]
+            if candidates is None:
+                candidates = list(self._by_kind.get(kind, {}).values())
+            for obj in candidates:
+                ns = obj.metadata.namespace
                 if namespace is not None and ns != namespace:
                     continue
                 if label_selector is not None:
@@ -161,7 +205,7 @@ class Store:
                         continue
                 if filter_fn is not None and not filter_fn(obj):
                     continue
-                out.append(serde.deep_copy(obj))
+                out.append(obj if not copy else serde.deep_copy(obj))
         out.sort(key=lambda o: (o.metadata.namespace, o.metadata.name))
         return out
 
@@ -198,6 +242,8 @@ class Store:
             if hasattr(obj, "status"):
                 obj.status = serde.deep_copy(old.status)
             self._objects[key] = obj
+            self._index_remove(key, self._by_kind.get(key[0], {}).get(key))
+            self._index_add(key, obj)
             self._by_kind.setdefault(key[0], {})[key] = obj
             events.append((MODIFIED, obj))
         self._dispatch(events)
@@ -220,6 +266,8 @@ class Store:
             stored.status = serde.deep_copy(obj.status)
             stored.metadata.resource_version = str(self._rv)
             self._objects[key] = stored
+            self._index_remove(key, self._by_kind.get(key[0], {}).get(key))
+            self._index_add(key, stored)
             self._by_kind.setdefault(key[0], {})[key] = stored
             events.append((MODIFIED, stored))
         self._dispatch(events)
@@ -270,6 +318,7 @@ class Store:
                     obj.metadata.resource_version = str(self._rv)
                     events.append((MODIFIED, obj))
                 elif not obj.metadata.finalizers:
+                    self._index_remove(key, obj)
                     del self._objects[key]
                     self._by_kind.get(key[0], {}).pop(key, None)
                     events.append((DELETED, obj))
@@ -339,6 +388,7 @@ class Store:
                 if FOREGROUND_FINALIZER in obj.metadata.finalizers and \
                         self._dependents_locked(obj.metadata.uid):
                     continue
+                self._index_remove(key, obj)
                 del self._objects[key]
                 self._by_kind.get(key[0], {}).pop(key, None)
                 events.append((DELETED, obj))

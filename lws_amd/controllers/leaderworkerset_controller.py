@@ -302,13 +302,13 @@ class LeaderWorkerSetReconciler:
         ns = lws.metadata.namespace
         leader_pods = self.store.list("Pod", ns, label_selector={
             lwsapi.SET_NAME_LABEL_KEY: lws.metadata.name,
-            lwsapi.WORKER_INDEX_LABEL_KEY: "0"})
+            lwsapi.WORKER_INDEX_LABEL_KEY: "0"}, copy=False)
         sorted_pods = sort_by_index(
             leader_pods,
             lambda p: int(p.metadata.labels[lwsapi.GROUP_INDEX_LABEL_KEY]),
             sts_replicas)
         sts_list = self.store.list("StatefulSet", ns, label_selector={
-            lwsapi.SET_NAME_LABEL_KEY: lws.metadata.name})
+            lwsapi.SET_NAME_LABEL_KEY: lws.metadata.name}, copy=False)
         sorted_sts = sort_by_index(
             sts_list,
             lambda s: int((s.metadata.labels or {}).get(
@@ -472,7 +472,13 @@ class LeaderWorkerSetReconciler:
         ns = lws.metadata.namespace
         leader_pods = self.store.list("Pod", ns, label_selector={
             lwsapi.SET_NAME_LABEL_KEY: lws.metadata.name,
-            lwsapi.WORKER_INDEX_LABEL_KEY: "0"})
+            lwsapi.WORKER_INDEX_LABEL_KEY: "0"}, copy=False)
+        # one indexed bulk fetch of the worker STSes instead of a deep-
+        # copying try_get per replica (read-only lister refs)
+        sts_by_name = {
+            o.metadata.name: o
+            for o in self.store.list("StatefulSet", ns, label_selector={
+                lwsapi.SET_NAME_LABEL_KEY: lws.metadata.name}, copy=False)}
         ready_count = updated_count = ready_non_burst = 0
         part_updated_non_burst = part_current_non_burst = part_updated_ready = 0
         no_worker_sts = lws.spec.leader_worker_template.size == 1
@@ -487,7 +493,7 @@ class LeaderWorkerSetReconciler:
                 continue
             sts = None
             if not no_worker_sts:
-                sts = self.store.try_get("StatefulSet", ns, pod.metadata.name)
+                sts = sts_by_name.get(pod.metadata.name)
                 if sts is None:
                     continue
             if index < lws_replicas and index >= lws_partition:
