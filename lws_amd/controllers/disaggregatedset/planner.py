@@ -258,8 +258,11 @@ def compute_all_steps(initial_old, target, config) -> list[UpdateStep]:
     n = len(initial_old)
     current_old = list(initial_old)
     current_new = [0] * n
-    max_steps = max([0] + [max(initial_old[i], target[i])
-                           for i in range(n)]) * 2 + 10
+    # every step strictly drains old or grows new in at least one role, so
+    # the worst case is bounded by the total unit changes across roles
+    # (a per-role max bound undercounts interleaved up/down sequences,
+    # e.g. [12,12]->[1,12] takes 35 steps)
+    max_steps = sum(initial_old) + sum(target) + 10
     steps = [UpdateStep(past=list(initial_old), new=[0] * n)]
     for _ in range(max_steps):
         nxt = compute_next_step(initial_old, current_old, current_new, target,

@@ -114,3 +114,15 @@ def test_planner_properties(n, data):
         assert steps[-1].past == [0] * n
         return
     check_rollout(initial_old, target, config)
+
+
+def test_interleaved_drain_scale_up_completes():
+    """Regression (hypothesis find): [12,12]->[1,12] with surge 1 takes 35
+    steps — the old per-role-max simulator bound cut it off one short."""
+    from lws_amd.controllers.disaggregatedset.planner import (
+        RollingUpdateConfig, compute_all_steps)
+
+    cfg = [RollingUpdateConfig(max_surge=1, max_unavailable=0)] * 2
+    steps = compute_all_steps([12, 12], [1, 12], cfg)
+    assert steps[-1].past == [0, 0]
+    assert steps[-1].new == [1, 12]
