@@ -1,7 +1,7 @@
 # lws-amd developer targets (reference Makefile analogue)
 PY ?= python3
 
-.PHONY: build test test-gpu bench smoke lint plan-steps manager clean
+.PHONY: build test test-integration test-e2e test-gpu bench bench-ds smoke lint plan-steps manager clean
 
 build:          ## build the gfx950 HIP kernel extension in-tree
 	$(PY) -m lws_amd.ops.build
@@ -9,11 +9,23 @@ build:          ## build the gfx950 HIP kernel extension in-tree
 test:           ## CPU test tier (control plane, engine reference path, gloo TP)
 	$(PY) -m pytest tests -q -m "not gpu"
 
+test-integration:  ## in-process cluster tier (fake-runtime agents; envtest analogue)
+	$(PY) -m pytest tests/test_lws_create.py tests/test_rolling_update.py \
+	    tests/test_disaggregatedset.py tests/test_gang_scheduling.py \
+	    tests/test_webhook_behaviors.py tests/test_events_and_resume.py -q
+
+test-e2e:       ## subprocess manager + HTTP client + lwsctl (kind-e2e analogue)
+	$(PY) -m pytest tests/test_manager_e2e.py tests/test_serving_server.py \
+	    tests/test_serving_tp_cpu.py -q
+
 test-gpu:       ## MI355X tier (kernel numerics vs fp32 reference, HIP engine)
 	$(PY) -m pytest tests -q -m gpu
 
 bench:          ## flagship benchmark (BASELINE.json metric), 1 GPU
 	$(PY) bench.py --gpus 1 --steps 3 --warmup 1
+
+bench-ds:       ## DisaggregatedSet lifecycle benchmark (config #4)
+	$(PY) scripts/bench_ds.py --model llama-3-8b --steps 3 --warmup 1
 
 smoke:
 	$(PY) -c "import __graft_entry__ as g; g.build(); g.smoke()"

@@ -91,10 +91,19 @@ def llama_tiny() -> LlamaConfig:
                        max_position=2048)
 
 
+def llama_tiny4() -> LlamaConfig:
+    """llama-tiny with 4 kv heads so TP=4 multi-process tests divide."""
+    return LlamaConfig(name="llama-tiny4", hidden_size=256,
+                       intermediate_size=512, num_layers=2, num_q_heads=8,
+                       num_kv_heads=4, head_dim=128, vocab_size=1024,
+                       max_position=2048)
+
+
 MODEL_PRESETS = {
     "llama-3-8b": llama3_8b,
     "llama-3-70b": llama3_70b,
     "llama-tiny": llama_tiny,
+    "llama-tiny4": llama_tiny4,
     "mixtral-8x7b": mixtral_8x7b,
     "mixtral-tiny": mixtral_tiny,
 }

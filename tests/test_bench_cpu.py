@@ -2,6 +2,7 @@
 with llama-tiny on the fp32 reference ops and gloo collectives."""
 import json
 import os
+import random
 import subprocess
 import sys
 
@@ -83,3 +84,49 @@ def test_bench_two_rank_gloo():
     assert out["n_gpus"] == 2
     assert out["config"]["parallelism"] == "tp2"
     assert out["value"] > 0
+
+
+def test_bench_four_rank_gloo():
+    """4-rank gloo dry-run of the driver's N=4 launch shape (TP=4 over
+    the collective runtime; llama-tiny4 divides 4 kv heads)."""
+    from tests.test_bench_cpu import _parse_json_line
+
+    for attempt in range(3):
+        port = random.randint(20000, 60000)
+        procs = []
+        for rank in range(4):
+            p_env = dict(os.environ)
+            p_env.update({"RANK": str(rank), "WORLD_SIZE": "4",
+                          "MASTER_ADDR": "127.0.0.1",
+                          "MASTER_PORT": str(port),
+                          "LOCAL_RANK": str(rank)})
+            cmd = [sys.executable, os.path.join(REPO, "bench.py"),
+                   "--device", "cpu", "--model", "llama-tiny4",
+                   "--kv-pages", "64", "--decode-batch", "4",
+                   "--prompt-len", "16", "--decode-steps", "2",
+                   "--gpus", "4", "--steps", "1", "--warmup", "0"]
+            procs.append(subprocess.Popen(cmd, stdout=subprocess.PIPE,
+                                          stderr=subprocess.PIPE, text=True,
+                                          env=p_env, cwd=REPO))
+        try:
+            outs = []
+            ok = True
+            for p in procs:
+                stdout, stderr = p.communicate(timeout=300)
+                if p.returncode != 0:
+                    ok = False
+                    last_err = stderr[-3000:]
+                outs.append(stdout)
+            if ok:
+                out = _parse_json_line(outs[0])
+                assert out["n_gpus"] == 4
+                assert out["config"]["parallelism"] == "tp4"
+                assert out["value"] > 0
+                return
+        except subprocess.TimeoutExpired:
+            last_err = "timeout"
+        finally:
+            for p in procs:
+                if p.poll() is None:
+                    p.kill()
+    raise AssertionError(f"4-rank bench failed 3 attempts: {last_err}")
