@@ -200,6 +200,12 @@ class Engine:
     # -- request management --------------------------------------------
     def add_request(self, prompt_ids: list[int],
                     sampling: Optional[SamplingParams] = None) -> int:
+        if not prompt_ids:
+            raise ValueError("prompt must be non-empty")
+        if len(prompt_ids) >= self.cfg.max_model_len:
+            raise ValueError(
+                f"prompt length {len(prompt_ids)} exceeds max_model_len "
+                f"{self.cfg.max_model_len}")
         sid = self._next_seq_id
         self._next_seq_id += 1
         sp = sampling or SamplingParams()

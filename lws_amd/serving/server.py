@@ -143,6 +143,13 @@ def build_app(loop: ServingLoop, model_name: str):
             vocab = loop.engine.model_cfg.vocab_size
             prompt = [b % vocab for b in prompt.encode()] or [0]
         max_tokens = int(body.get("max_tokens", 16))
+        if max_tokens < 1:
+            raise HTTPException(400, "max_tokens must be >= 1")
+        limit = loop.engine.model_cfg.max_position
+        if len(prompt) + max_tokens > limit:
+            raise HTTPException(
+                400, f"prompt+max_tokens {len(prompt) + max_tokens} exceeds "
+                     f"model context {limit}")
         sp = SamplingParams(
             temperature=float(body.get("temperature", 0.0)),
             top_p=float(body.get("top_p", 1.0)),
