@@ -276,6 +276,18 @@ class LlamaForCausalLM:
         n += self.embed.numel() + self.final_norm.numel() + self.lm_head.numel
         return n
 
+    def quantize_fp8(self) -> None:
+        """Opt-in fp8-weights mode: quantize every dense projection
+        (qkv/o/gate_up/down/lm_head) to OCP e4m3 with per-tensor scales;
+        embed/norms/router stay bf16.  MoE expert stacks keep bf16 (the
+        grouped skinny kernel is bf16) — dense models only for now."""
+        assert self.cfg.num_experts == 0, \
+            "fp8 weight mode supports dense models only"
+        for layer in self.layers:
+            for lin in (layer.qkv, layer.o, layer.gate_up, layer.down):
+                lin.quantize_fp8()
+        self.lm_head.quantize_fp8()
+
     # -- kv cache shape -------------------------------------------------
     def kv_cache_spec(self) -> tuple[int, int]:
         """(num_layers, kv_heads_per_rank)"""

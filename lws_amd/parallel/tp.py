@@ -99,12 +99,36 @@ class ShardedLinear:
             assert in_features % tp_world == 0
             shape = (out_features, in_features // tp_world)
         self.weight = torch.empty(shape, device=device, dtype=dtype)
+        self.weight_fp8: Optional[torch.Tensor] = None
+        self.scale_fp8: Optional[torch.Tensor] = None
 
     def materialize(self, generator: Optional[torch.Generator] = None,
                     std: float = 0.02) -> None:
         self.weight.normal_(0.0, std, generator=generator)
 
+    def quantize_fp8(self) -> None:
+        """Opt-in fp8-weights serving mode (OCP e4m3, per-tensor scale):
+        halves the decode weight-read; activations are quantized
+        per-call (W8A8 _scaled_mm, bf16 out).  The bf16 weights are
+        freed — this linear then serves fp8 only."""
+        w = self.weight
+        self.scale_fp8 = (w.abs().amax().float() / 448.0).clamp(min=1e-8)
+        self.weight_fp8 = (w.float() / self.scale_fp8).clamp(
+            -448.0, 448.0).to(torch.float8_e4m3fn)
+        self.weight = None
+
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        if self.weight_fp8 is not None:
+            M = x.size(0)
+            pad = (-M) % 16
+            if pad:
+                x = torch.nn.functional.pad(x, (0, 0, 0, pad))
+            xs = (x.abs().amax().float() / 448.0).clamp(min=1e-8)
+            x8 = (x.float() / xs).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+            out = torch._scaled_mm(x8, self.weight_fp8.t(), scale_a=xs,
+                                   scale_b=self.scale_fp8,
+                                   out_dtype=torch.bfloat16)
+            return out[:M] if pad else out
         if x.is_cuda and x.dim() == 2 and x.size(0) <= 32 and \
                 x.size(1) % 128 == 0 and self.weight.numel() <= 32 * 1024 * 1024:
             # decode-shape path for small weight shards (<=64 MB bf16, the
@@ -119,4 +143,5 @@ class ShardedLinear:
 
     @property
     def numel(self) -> int:
-        return self.weight.numel()
+        w = self.weight if self.weight is not None else self.weight_fp8
+        return w.numel()

@@ -46,9 +46,13 @@ class ShardHost:
 
     def build(self, spec: dict) -> dict:
         self.teardown()
+        import os
         cfg = EngineConfig(model=spec["model"],
                            kv_pages=int(spec.get("kv_pages", 128)),
                            seed=int(spec.get("seed", 0)),
+                           weight_dtype=spec.get(
+                               "weight_dtype",
+                               os.environ.get("LWS_AMD_WEIGHT_DTYPE", "bf16")),
                            device=self.device, tp_rank=self.rank,
                            tp_world=self.world)
         self.engine = Engine(cfg)
