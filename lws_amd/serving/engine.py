@@ -362,20 +362,25 @@ class Engine:
         kv_row_idx = kv_starts = q_offsets = None
         if chunked:
             # gather index over the FULL context (prefix + chunk) per
-            # (token, head): flat cache row (page*Hkv + h)*page + off
+            # (token, head): flat cache row (page*Hkv + h)*page + off.
+            # Vectorized — a python loop over tokens x heads costs ~10 ms
+            # at an 8k-token chunk.
             hkv = self.kv_caches[0][0].size(1)
-            rows, kv_starts, q_offsets = [], [0], []
+            kv_starts, q_offsets, bases = [0], [], []
             for s, take in sel:
                 end = s.num_cached + take
-                for t in range(end):
-                    base = s.pages[t // PAGE_SIZE] * hkv * PAGE_SIZE + \
-                        t % PAGE_SIZE
-                    for h in range(hkv):
-                        rows.append(base + h * PAGE_SIZE)
+                pages = torch.tensor(
+                    s.pages[:(end + PAGE_SIZE - 1) // PAGE_SIZE],
+                    dtype=torch.int64)
+                t = torch.arange(end, dtype=torch.int64)
+                bases.append(pages[t // PAGE_SIZE] * (hkv * PAGE_SIZE)
+                             + t % PAGE_SIZE)
                 kv_starts.append(kv_starts[-1] + end)
                 q_offsets.append(s.num_cached)
-            kv_row_idx = torch.tensor(rows, device=self.device,
-                                      dtype=torch.int64)
+            base = torch.cat(bases)                       # [Tkv]
+            h_off = torch.arange(hkv, dtype=torch.int64) * PAGE_SIZE
+            kv_row_idx = (base[:, None] + h_off[None, :]).reshape(-1) \
+                .to(self.device)
         batch = PrefillBatch(
             input_ids=torch.tensor(ids, device=self.device, dtype=torch.long),
             positions=torch.tensor(pos, device=self.device, dtype=torch.int32),
