@@ -239,12 +239,12 @@ class Engine:
         """Flat cache-row index per (token, head): the KV caches are
         [pages, Hkv, page, D], so row (page*Hkv + h)*page_size + off."""
         hkv = self.kv_caches[0][0].size(1)
-        rows = []
-        for t in range(upto):
-            base = seq.pages[t // PAGE_SIZE] * hkv * PAGE_SIZE + t % PAGE_SIZE
-            for h in range(hkv):
-                rows.append(base + h * PAGE_SIZE)
-        return torch.tensor(rows, device=self.device, dtype=torch.int64)
+        pages = torch.tensor(seq.pages[:(upto + PAGE_SIZE - 1) // PAGE_SIZE],
+                             dtype=torch.int64)
+        t = torch.arange(upto, dtype=torch.int64)
+        base = pages[t // PAGE_SIZE] * (hkv * PAGE_SIZE) + t % PAGE_SIZE
+        h_off = torch.arange(hkv, dtype=torch.int64) * PAGE_SIZE
+        return (base[:, None] + h_off[None, :]).reshape(-1).to(self.device)
 
     def export_kv(self, seq_id: int) -> tuple[list[int], int, list]:
         """Extract a sequence's cached KV for handoff to another engine
