@@ -14,6 +14,7 @@ sequences finish.
 """
 from __future__ import annotations
 
+import queue
 import threading
 import time
 import uuid
@@ -31,6 +32,8 @@ class _Pending:
     max_tokens: int
     future: Future = field(default_factory=Future)
     created: float = field(default_factory=time.time)
+    stream_q: Optional[queue.Queue] = None   # per-token streaming sink
+    sent: int = 0                            # tokens already streamed
 
 
 class ServingLoop:
@@ -50,14 +53,16 @@ class ServingLoop:
         return self
 
     def submit(self, prompt_ids: list[int], max_tokens: int,
-               sampling: Optional[SamplingParams] = None) -> Future:
+               sampling: Optional[SamplingParams] = None,
+               stream: bool = False) -> "_Pending":
         with self._lock:
             sid = self.engine.add_request(list(prompt_ids), sampling)
             p = _Pending(seq_id=sid, prompt_len=len(prompt_ids),
-                         max_tokens=max_tokens)
+                         max_tokens=max_tokens,
+                         stream_q=queue.Queue() if stream else None)
             self._pending[sid] = p
             self.stats["requests"] += 1
-        return p.future
+        return p
 
     def _run(self) -> None:
         while not self._stop.is_set():
@@ -73,6 +78,8 @@ class ServingLoop:
                             self.engine.finish(p.seq_id)
                             if not p.future.done():
                                 p.future.set_exception(e)
+                            if p.stream_q is not None:
+                                p.stream_q.put(None)
                         self._pending.clear()
                         continue
                     self.stats["steps"] += 1
@@ -85,11 +92,20 @@ class ServingLoop:
                             finished.append(sid)
                             continue
                         produced = len(seq.token_ids) - p.prompt_len
+                        if p.stream_q is not None and produced > p.sent:
+                            for tok in seq.token_ids[
+                                    p.prompt_len + p.sent:
+                                    p.prompt_len + min(produced,
+                                                       p.max_tokens)]:
+                                p.stream_q.put(tok)
+                            p.sent = min(produced, p.max_tokens)
                         if produced >= p.max_tokens or seq.finished:
                             tokens = seq.token_ids[
                                 p.prompt_len:p.prompt_len + p.max_tokens]
                             self.engine.finish(sid)
                             p.future.set_result(tokens)
+                            if p.stream_q is not None:
+                                p.stream_q.put(None)     # end-of-stream
                             finished.append(sid)
                     for sid in finished:
                         self._pending.pop(sid, None)
@@ -155,8 +171,35 @@ def build_app(loop: ServingLoop, model_name: str):
             top_p=float(body.get("top_p", 1.0)),
             top_k=int(body.get("top_k", 0)),
             seed=(int(body["seed"]) if "seed" in body else None))
-        fut = loop.submit(prompt, max_tokens, sp)
-        tokens = fut.result(timeout=float(body.get("timeout", 300)))
+        if body.get("stream"):
+            import json as _json
+
+            from fastapi.responses import StreamingResponse
+
+            pend = loop.submit(prompt, max_tokens, sp, stream=True)
+            rid = f"cmpl-{uuid.uuid4().hex[:12]}"
+
+            def sse():
+                deadline = time.time() + float(body.get("timeout", 300))
+                while time.time() < deadline:
+                    try:
+                        tok = pend.stream_q.get(timeout=1.0)
+                    except queue.Empty:
+                        if pend.future.done():
+                            break
+                        continue
+                    if tok is None:
+                        break
+                    chunk = {"id": rid, "object": "text_completion.chunk",
+                             "model": model_name,
+                             "choices": [{"index": 0, "token_ids": [tok],
+                                          "text": str(tok)}]}
+                    yield f"data: {_json.dumps(chunk)}\n\n"
+                yield "data: [DONE]\n\n"
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        pend = loop.submit(prompt, max_tokens, sp)
+        tokens = pend.future.result(timeout=float(body.get("timeout", 300)))
         return {
             "id": f"cmpl-{uuid.uuid4().hex[:12]}",
             "object": "text_completion",

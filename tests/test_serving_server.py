@@ -17,8 +17,8 @@ def _make_loop():
 def test_serving_loop_completes_requests():
     loop = _make_loop()
     try:
-        f1 = loop.submit([1, 2, 3], max_tokens=4)
-        f2 = loop.submit([9, 8, 7, 6], max_tokens=2)
+        f1 = loop.submit([1, 2, 3], max_tokens=4).future
+        f2 = loop.submit([9, 8, 7, 6], max_tokens=2).future
         t1 = f1.result(timeout=60)
         t2 = f2.result(timeout=60)
         assert len(t1) == 4 and len(t2) == 2
@@ -30,7 +30,7 @@ def test_serving_loop_completes_requests():
 def test_serving_loop_concurrent_submit():
     loop = _make_loop()
     try:
-        futs = [loop.submit([i, i + 1], max_tokens=3) for i in range(6)]
+        futs = [loop.submit([i, i + 1], max_tokens=3).future for i in range(6)]
         outs = [f.result(timeout=120) for f in futs]
         assert all(len(o) == 3 for o in outs)
     finally:
@@ -75,11 +75,44 @@ def test_serving_loop_fails_requests_on_engine_error():
     loop = ServingLoop(eng).start()
     try:
         # 3 pages usable (page 0 reserved); this prompt needs 5 pages
-        f = loop.submit(list(range(70)), max_tokens=2)
+        f = loop.submit(list(range(70)), max_tokens=2).future
         with pytest.raises(RuntimeError, match="KV cache exhausted"):
             f.result(timeout=30)
         # small request still succeeds afterwards
-        f2 = loop.submit([1, 2, 3], max_tokens=2)
+        f2 = loop.submit([1, 2, 3], max_tokens=2).future
         assert len(f2.result(timeout=60)) == 2
+    finally:
+        loop.stop()
+
+
+def test_streaming_completions_endpoint():
+    """SSE streaming: per-token chunks then [DONE]."""
+    from fastapi.testclient import TestClient
+
+    from lws_amd.serving.engine import Engine, EngineConfig
+    from lws_amd.serving.server import ServingLoop, build_app
+
+    eng = Engine(EngineConfig(model="llama-tiny", device="cpu", kv_pages=64,
+                              seed=7))
+    eng.load()
+    loop = ServingLoop(eng).start()
+    try:
+        app = build_app(loop, "llama-tiny")
+        client = TestClient(app)
+        with client.stream("POST", "/v1/completions",
+                           json={"prompt": [3, 1, 4], "max_tokens": 4,
+                                 "stream": True}) as r:
+            assert r.status_code == 200
+            body = "".join(r.iter_text())
+        chunks = [l for l in body.split("\n") if l.startswith("data: ")]
+        assert chunks[-1] == "data: [DONE]"
+        import json as _json
+        toks = [_json.loads(c[6:])["choices"][0]["token_ids"][0]
+                for c in chunks[:-1]]
+        assert len(toks) == 4
+        # matches the non-streaming result
+        want = client.post("/v1/completions",
+                           json={"prompt": [3, 1, 4], "max_tokens": 4})
+        assert want.json()["choices"][0]["token_ids"] == toks
     finally:
         loop.stop()
