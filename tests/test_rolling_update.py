@@ -95,7 +95,7 @@ def test_rolling_update_replaces_all_groups(cluster):
         c = cluster.get_lws("default", "roll")
         cond = lws_condition(c, "UpdateInProgress")
         return c if cond is not None and cond.status == "True" else None
-    wait_for(updating, desc="UpdateInProgress", timeout=10)
+    wait_for(updating, desc="UpdateInProgress", timeout=30)
 
     def done():
         c = cluster.get_lws("default", "roll")
@@ -106,12 +106,18 @@ def test_rolling_update_replaces_all_groups(cluster):
                 else None)
     wait_for(done, desc="rolling update complete", timeout=60)
 
-    # every group pod replaced, new image everywhere
-    pods_after = cluster.store.list("Pod", "default")
-    assert len(pods_after) == 6
-    for p in pods_after:
-        assert p.metadata.uid not in pods_before.values()
-        assert p.spec.containers[0].image == "engine:v2"
+    # every group pod replaced, new image everywhere (an old pod can
+    # still be terminating the instant Available flips — poll)
+    def replaced():
+        live = [p for p in cluster.store.list("Pod", "default")
+                if p.metadata.deletion_timestamp is None]
+        if len(live) != 6:
+            return None
+        ok = all(p.metadata.uid not in pods_before.values()
+                 and p.spec.containers[0].image == "engine:v2"
+                 for p in live)
+        return ok or None
+    wait_for(replaced, desc="all pods replaced on v2", timeout=30)
 
     # revisions truncated to the current one (happens right after the final
     # status write in the same reconcile — poll briefly)
