@@ -1,28 +1,36 @@
 #include "hip/hip_runtime.h"
-// Split-K MFMA "skinny" GEMM for decode-shape projections on gfx950. (v3)
+// Split-K MFMA "skinny" GEMM for decode-shape projections on gfx950. (v5)
 //
-// Motivation (profiles/r01_decode8b_kernel_stats.md): at decode batch
-// M<=32, hipBLASLt's heuristic tiles underfill the 256-CU chip and read
-// cold weights at ~1 TB/s.  This kernel splits K so the launch fills the
-// chip and streams W at HBM rate.
+// Motivation (profiles/r01_skinny_dispatch.md): at decode batch M<=32,
+// hipBLASLt's heuristic tiles underfill the 256-CU chip on small TP
+// shards.  This kernel splits K so the launch fills the chip and streams
+// W at HBM rate; the measured dispatch policy in parallel/tp.py sends
+// shards <=32M elements here and bigger shapes to hipBLASLt.
 //
 //   out[M, N] = x[M, K] @ W[N, K]^T      (bf16 in, fp32 accumulate)
+//   grouped:  out[E, M, N] = x[E, M, K] @ W[E, N, K]^T  (MoE experts,
+//   one launch streams every expert — blockIdx.z selects the expert)
 //
-// v3 design (CDNA4 guide §5 + technique catalog):
+// v5 design (CDNA4 guide §5 + technique catalog):
 //  - async global->LDS staging via __builtin_amdgcn_global_load_lds
 //    (16 B/lane direct DMA, no VGPR round trip — Common-mistake #1)
-//  - double-buffered K sub-slices: next tile's loads issue BEFORE the
-//    current tile's MFMA loop, the trailing __syncthreads drains them
-//    (minimum 2-phase pattern, guide §5.5 T3 recipe)
-//  - LDS destination of global_load_lds is linear (wave-uniform base +
-//    lane*16), so bank-conflict avoidance uses the XOR swizzle applied to
-//    BOTH the per-lane global source address and the ds_read offset —
-//    same involution on both sides (guide ERRATA #21 / T2)
+//  - 3-deep pipeline, 2 sub-slices staged ahead, counted s_waitcnt
+//    vmcnt(N) + raw s_barrier at the loop head — staging is never
+//    force-drained (guide §5 K-loop / §5.5 T3+T4); 72 KB LDS keeps 2
+//    workgroups resident per CU for the latency-bound cold regime
+//  - LDS fragment reads via inline-asm ds_read_b128: a compiled read
+//    aliasing the LDS-DMA buffers gets a compiler-inserted vmcnt(0)
+//    that re-serializes the pipeline (verified in ISA dumps); fragment
+//    registers are threaded through an explicit s_waitcnt lgkmcnt(0)
+//    so the scheduler cannot hoist MFMAs above the wait
+//  - XOR swizzle applied to BOTH the per-lane global source address and
+//    the ds_read offset — same involution on both sides (ERRATA #21/T2)
 //  - one workgroup = 4 waves; each wave owns a 16-wide n-tile (64 n per
 //    block) and a K-slice; per 32-k step each wave issues ceil(M/16)
 //    mfma_f32_16x16x32_bf16
 //  - deterministic split-K: per-slice fp32 partial planes + a finalize
-//    reduce kernel (no atomics)
+//    reduce kernel (no atomics); unsplit launches (grid_y==1, incl. the
+//    grouped MoE form) write bf16 directly — no workspace round trip
 //
 // Fragment layouts (verified on MI355X by tests/test_kernels_gpu.py
 // numerics with asymmetric inputs — guide §3 rule):
