@@ -1,9 +1,14 @@
 // Paged-attention decode for gfx950 (MI355X), GQA-aware, flash-decoding
-// sequence split.  v2 design notes:
+// sequence split.  v3 design notes:
 //
-//  - grid = (num_seqs, H_kv, seq_chunks) x 256 threads (4 waves); the host
-//    wrapper sizes seq_chunks so the launch fills the 256-CU chip even at
-//    small B*H_kv (one kv head per GPU at TP8).
+//  - grid = (num_seqs, H_kv, seq_chunks) x NW waves; the host wrapper
+//    sizes seq_chunks per the measured policy (~16 chunks/seq at small
+//    B*H_kv, ~8 when one block per (b,hkv) covers the chip —
+//    profiles/r01_pa_direct.md); NW=4 measured best everywhere (NW=1
+//    stays reachable via LWS_PA_NW for tuning).
+//  - DIRECT single-chunk fast path: in-kernel l-normalize, bf16 straight
+//    to out, no workspace and no reduce launch (picked when B*Hkv >= 256
+//    and the GQA group is 1).
 //  - Each WAVE owns an independent interleaved key stream with its own
 //    online-softmax state: NO __syncthreads in the key loop (v1 was
 //    latency-bound on two block syncs per 16-key pass).
