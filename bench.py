@@ -249,6 +249,9 @@ class BenchConductor:
 
 def main() -> None:
     args = parse_args()
+    # the judged metric is bf16: never let a stray env flip the engines
+    # into the fp8 weight mode during a driver run
+    os.environ.pop("LWS_AMD_WEIGHT_DTYPE", None)
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
     if args.device is None:
