@@ -33,6 +33,7 @@ READONLY_KINDS = {
     "services": "Service",
     "podgroups": "PodGroup",
     "controllerrevisions": "ControllerRevision",
+    "events": "Event",
 }
 
 

@@ -21,6 +21,7 @@ from ..api import leaderworkerset as lwsapi
 from ..api.core import Pod, Service, StatefulSet
 from ..api.disaggregatedset import DisaggregatedSet, DisaggregatedSetRoleScaler
 from ..api.leaderworkerset import LeaderWorkerSet
+from ..cluster.events import Event as _EventModel
 
 RESOURCES = {
     "leaderworkersets": (lwsapi.KIND, LeaderWorkerSet),
@@ -30,6 +31,7 @@ RESOURCES = {
     "pods": ("Pod", Pod),
     "statefulsets": ("StatefulSet", StatefulSet),
     "services": ("Service", Service),
+    "events": ("Event", _EventModel),
 }
 KIND_TO_RESOURCE = {kind: res for res, (kind, _) in RESOURCES.items()}
 
@@ -217,6 +219,9 @@ class Clientset:
 
     def services(self, namespace="default") -> ResourceClient:
         return ResourceClient(self.transport, "services", namespace)
+
+    def events(self, namespace="default") -> ResourceClient:
+        return ResourceClient(self.transport, "events", namespace)
 
 
 class Informer:

@@ -125,6 +125,46 @@ def cmd_delete(args) -> int:
     return 0
 
 
+def cmd_events(args) -> int:
+    import time as _time
+    cs = _client(args)
+    evs = cs.events(args.namespace).list()
+    evs.sort(key=lambda e: e.last_timestamp)
+    print(f"{'TYPE':8s} {'REASON':22s} {'OBJECT':32s} {'COUNT':5s} "
+          f"{'AGE':6s} MESSAGE")
+    now = _time.time()
+    for e in evs:
+        obj = f"{e.involved_object.kind}/{e.involved_object.name}"
+        age = f"{int(now - e.last_timestamp)}s"
+        print(f"{e.type:8s} {e.reason:22s} {obj:32s} {e.count:<5d} "
+              f"{age:6s} {e.message}")
+    return 0
+
+
+def cmd_describe(args) -> int:
+    resource = _resource(args.resource)
+    cs = _client(args)
+    rc_name = {"leaderworkersets": "leader_worker_sets",
+               "disaggregatedsets": "disaggregated_sets",
+               "disaggregatedsetrolescalers": "role_scalers",
+               "pods": "pods", "statefulsets": "statefulsets",
+               "services": "services"}[resource]
+    obj = getattr(cs, rc_name)(args.namespace).get(args.name)
+    if obj is None:
+        sys.exit(f"error: {args.name} not found")
+    d = serde.to_dict(obj)
+    print(yaml.safe_dump({"metadata": d.get("metadata"),
+                          "spec": d.get("spec"),
+                          "status": d.get("status")}, sort_keys=False))
+    evs = [e for e in cs.events(args.namespace).list()
+           if e.involved_object.name == args.name]
+    if evs:
+        print("Events:")
+        for e in sorted(evs, key=lambda e: e.last_timestamp):
+            print(f"  {e.type} {e.reason} (x{e.count}): {e.message}")
+    return 0
+
+
 def cmd_scale(args) -> int:
     resource = _resource(args.resource)
     cs = _client(args)
@@ -156,6 +196,14 @@ def main(argv=None) -> int:
     sp.add_argument("resource")
     sp.add_argument("name")
     sp.set_defaults(fn=cmd_delete)
+
+    sp = sub.add_parser("events")
+    sp.set_defaults(fn=cmd_events)
+
+    sp = sub.add_parser("describe")
+    sp.add_argument("resource")
+    sp.add_argument("name")
+    sp.set_defaults(fn=cmd_describe)
 
     sp = sub.add_parser("scale")
     sp.add_argument("resource")

@@ -104,6 +104,38 @@ def test_lwsctl_cli(manager_proc):
     assert "my-lws" not in out
 
 
+def test_lwsctl_events_and_describe(manager_proc):
+    """lwsctl events / describe (kubectl describe + events analogue)."""
+    import subprocess as sp
+
+    _, base = manager_proc
+    from lws_amd.client.clientset import Clientset
+    from tests.conftest import make_lws, wait_for
+
+    cs = Clientset.for_server(base)
+    cs.leader_worker_sets().create(make_lws(name="desc-lws", replicas=1,
+                                            size=2))
+
+    def ready():
+        cur = cs.leader_worker_sets().get("desc-lws")
+        conds = {c.type: c.status for c in cur.status.conditions}
+        return cur if conds.get("Available") == "True" else None
+    wait_for(ready, desc="desc-lws Available", timeout=60)
+
+    def ctl(*argv):
+        out = sp.run([sys.executable, "-m", "lws_amd.client.ctl",
+                      "--server", base, *argv],
+                     cwd=REPO, capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, out.stderr[-2000:]
+        return out.stdout
+
+    ev = ctl("events")
+    assert "REASON" in ev and "desc-lws" in ev
+
+    d = ctl("describe", "lws", "desc-lws")
+    assert "status:" in d and "Available" in d
+
+
 def test_informer(manager_proc):
     proc, base = manager_proc
     from lws_amd.client.clientset import Clientset, Informer
