@@ -158,8 +158,16 @@ def test_recreate_group_after_start_gate():
         if still_pending:
             # the gate held: restart while pending must not recreate
             assert now == uids, "group must not recreate while pods pending"
-        uids = now
         _wait_available(c, "ags", timeout=30)
+        # re-snapshot from the settled state: the pre-Available snapshot
+        # can catch a pod mid-recreate (1 or 3 entries) and then the
+        # recreate check below could never match
+        pods = wait_for(
+            lambda: (lambda ps: ps if len(ps) == 2 and
+                     all(p.metadata.deletion_timestamp is None for p in ps)
+                     else None)(c.store.list("Pod", "default")),
+            desc="2 settled pods", timeout=30)
+        uids = {p.metadata.name: p.metadata.uid for p in pods}
 
         # once started, a restart recreates the group
         bump_restart("ags-0-1")
