@@ -241,21 +241,36 @@ class LlamaLayer:
 
 class LlamaForCausalLM:
     def __init__(self, cfg: LlamaConfig, tp_rank: int = 0, tp_world: int = 1,
-                 device: str = "cpu", dtype=torch.bfloat16):
+                 device: str = "cpu", dtype=torch.bfloat16,
+                 pp_rank: int = 0, pp_world: int = 1):
         self.cfg = cfg
         self.tp_rank = tp_rank
         self.tp_world = tp_world
+        self.pp_rank = pp_rank
+        self.pp_world = pp_world
+        assert tp_world == 1 or pp_world == 1, \
+            "combined TP x PP grids are not supported"
         self.device = torch.device(device)
         self.dtype = dtype
         self._ops = _Ops(self.device)
+        # contiguous layer slice for this pipeline stage (even split,
+        # remainder to the early stages)
+        L, base, rem = cfg.num_layers, cfg.num_layers // pp_world, \
+            cfg.num_layers % pp_world
+        sizes = [base + (1 if r < rem else 0) for r in range(pp_world)]
+        self.layer_offset = sum(sizes[:pp_rank])
+        self.pp_first = pp_rank == 0
+        self.pp_last = pp_rank == pp_world - 1
         self.layers = [LlamaLayer(cfg, tp_rank, tp_world, self.device, dtype)
-                       for _ in range(cfg.num_layers)]
+                       for _ in range(sizes[pp_rank])]
         self.embed = torch.empty(cfg.vocab_size, cfg.hidden_size,
-                                 device=self.device, dtype=dtype)
+                                 device=self.device, dtype=dtype) \
+            if self.pp_first else None
         self.final_norm = torch.empty(cfg.hidden_size, device=self.device,
-                                      dtype=dtype)
+                                      dtype=dtype) if self.pp_last else None
         self.lm_head = ShardedLinear(cfg.vocab_size, cfg.hidden_size, 0,
-                                     tp_rank, tp_world, self.device, dtype)
+                                     tp_rank, tp_world, self.device, dtype) \
+            if self.pp_last else None
         from lws_amd.ops import build_rope_table
         self.rope_table = build_rope_table(cfg.max_position, cfg.head_dim,
                                            cfg.rope_theta, device=self.device)
@@ -263,17 +278,27 @@ class LlamaForCausalLM:
 
     # -- weights --------------------------------------------------------
     def materialize(self, seed: int = 0) -> int:
-        """Random-init all shards on-device; returns local param count."""
-        gen = torch.Generator(device=self.device)
-        gen.manual_seed(seed + self.tp_rank)
+        """Random-init all shards on-device; returns local param count.
+        Seeding is per (global layer, tp_rank) so a PP stage reproduces
+        exactly the layer weights a single-process model would hold for
+        its slice (the PP==single equivalence tests rely on this)."""
+        def gen_for(tag: int) -> torch.Generator:
+            g = torch.Generator(device=self.device)
+            g.manual_seed(seed * 1000003 + tag * 7919 + self.tp_rank)
+            return g
         n = 0
-        for layer in self.layers:
-            layer.materialize(gen)
+        for i, layer in enumerate(self.layers):
+            layer.materialize(gen_for(self.layer_offset + i))
             n += layer.param_count
-        self.embed.normal_(0.0, 0.02, generator=gen)
-        self.final_norm.fill_(1.0)
-        self.lm_head.materialize(gen)
-        n += self.embed.numel() + self.final_norm.numel() + self.lm_head.numel
+        if self.embed is not None:
+            self.embed.normal_(0.0, 0.02, generator=gen_for(500009))
+            n += self.embed.numel()
+        if self.final_norm is not None:
+            self.final_norm.fill_(1.0)
+            n += self.final_norm.numel()
+        if self.lm_head is not None:
+            self.lm_head.materialize(gen_for(600011))
+            n += self.lm_head.numel
         return n
 
     def quantize_fp8(self) -> None:
@@ -290,8 +315,9 @@ class LlamaForCausalLM:
 
     # -- kv cache shape -------------------------------------------------
     def kv_cache_spec(self) -> tuple[int, int]:
-        """(num_layers, kv_heads_per_rank)"""
-        return self.cfg.num_layers, self.layers[0].hkv
+        """(local_layers, kv_heads_per_rank) — a PP stage pools KV only
+        for its own layer slice."""
+        return len(self.layers), self.layers[0].hkv
 
     # -- core ops (GPU: HIP kernels; CPU: fp32 reference) ---------------
     def _rmsnorm(self, x, w):
@@ -352,8 +378,16 @@ class LlamaForCausalLM:
     def forward_prefill(self, batch: PrefillBatch, kv_caches) -> torch.Tensor:
         """Returns hidden states of the LAST token of each sequence [B, H]."""
         cfg = self.cfg
-        x = self.embed[batch.input_ids]           # [T, H] (replicated embed)
-        residual = None
+        if self.pp_first:
+            x = self.embed[batch.input_ids]       # [T, H] (replicated embed)
+            residual = None
+        else:
+            # inter-layer state (x, residual) from the previous stage
+            from ..parallel import pp as ppmod
+            T = batch.input_ids.size(0)
+            st = ppmod.recv_stage((2, T, cfg.hidden_size), self.dtype,
+                                  self.device, self.pp_rank - 1)
+            x, residual = st[0], st[1]
         for li, layer in enumerate(self.layers):
             if residual is None:
                 residual = x
@@ -372,6 +406,12 @@ class LlamaForCausalLM:
             # MLP
             h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)
             x = all_reduce(self._mlp(layer, h))
+        B = len(batch.seq_starts) - 1
+        if not self.pp_last:
+            from ..parallel import pp as ppmod
+            ppmod.send_stage(torch.stack([x, residual]), self.pp_rank + 1)
+            return torch.zeros(B, self.cfg.hidden_size, device=self.device,
+                               dtype=self.dtype)
         h, _ = self._fused_add_rmsnorm(x, residual, self.final_norm)
         last = torch.tensor([s - 1 for s in batch.seq_starts[1:]],
                             device=h.device, dtype=torch.long)
@@ -426,8 +466,15 @@ class LlamaForCausalLM:
     def forward_decode(self, batch: DecodeBatch, kv_caches) -> torch.Tensor:
         """One token per sequence: returns hidden states [B, H]."""
         cfg = self.cfg
-        x = self.embed[batch.input_ids]           # [B, H]
-        residual = None
+        if self.pp_first:
+            x = self.embed[batch.input_ids]       # [B, H]
+            residual = None
+        else:
+            from ..parallel import pp as ppmod
+            B0 = batch.input_ids.size(0)
+            st = ppmod.recv_stage((2, B0, cfg.hidden_size), self.dtype,
+                                  self.device, self.pp_rank - 1)
+            x, residual = st[0], st[1]
         for li, layer in enumerate(self.layers):
             if residual is None:
                 residual = x
@@ -452,6 +499,10 @@ class LlamaForCausalLM:
             x = all_reduce(o)
             h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)
             x = all_reduce(self._mlp(layer, h))
+        if not self.pp_last:
+            from ..parallel import pp as ppmod
+            ppmod.send_stage(torch.stack([x, residual]), self.pp_rank + 1)
+            return torch.zeros_like(x)
         h, _ = self._fused_add_rmsnorm(x, residual, self.final_norm)
         return h
 
@@ -506,6 +557,11 @@ class LlamaForCausalLM:
         return out
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        """[N, H] -> [N, vocab] (column-sharded lm_head + all-gather)."""
+        """[N, H] -> [N, vocab] (column-sharded lm_head + all-gather).
+        Non-last PP stages return zeros — their sampled tokens are
+        overwritten by the last stage's broadcast in the engine."""
+        if self.lm_head is None:
+            return torch.zeros(hidden.size(0), self.cfg.vocab_size,
+                               device=hidden.device, dtype=hidden.dtype)
         local = self.lm_head(hidden)
         return all_gather_cat(local, dim=-1)

@@ -33,6 +33,8 @@ class EngineConfig:
     device: str = "cpu"
     tp_rank: int = 0
     tp_world: int = 1
+    pp_rank: int = 0                 # pipeline stage (TP x PP not combined)
+    pp_world: int = 1
 
 
 @dataclass
@@ -95,7 +97,9 @@ class Engine:
         self.model_cfg = mc
         self.model = LlamaForCausalLM(mc, tp_rank=cfg.tp_rank,
                                       tp_world=cfg.tp_world,
-                                      device=cfg.device)
+                                      device=cfg.device,
+                                      pp_rank=cfg.pp_rank,
+                                      pp_world=cfg.pp_world)
         self.device = self.model.device
         self.kv_caches: list[tuple[torch.Tensor, torch.Tensor]] = []
         self.allocator = BlockAllocator(cfg.kv_pages)
@@ -108,7 +112,7 @@ class Engine:
         # hipGraph capture of RCCL collectives is untested multi-rank on
         # this stack; default graphs to TP=1 only (LWS_AMD_GRAPHS=1 forces
         # them on for TP>1, LWS_AMD_NO_GRAPHS=1 disables everywhere)
-        graphs_ok = (cfg.tp_world == 1
+        graphs_ok = ((cfg.tp_world == 1 and cfg.pp_world == 1)
                      or os.environ.get("LWS_AMD_GRAPHS", "0") == "1")
         self.use_graphs = (self.device.type == "cuda" and graphs_ok
                            and os.environ.get("LWS_AMD_NO_GRAPHS", "0") != "1")
@@ -328,6 +332,14 @@ class Engine:
                 out[i] = torch.multinomial(probs, 1, generator=s.generator)
         return out
 
+    def _pp_sync_tokens(self, tokens: torch.Tensor) -> torch.Tensor:
+        """PP: only the last stage holds real logits — broadcast its
+        sampled tokens so every stage appends identical sequences."""
+        if self.cfg.pp_world <= 1:
+            return tokens
+        from ..parallel import pp as ppmod
+        return ppmod.broadcast_tokens(tokens, src=self.cfg.pp_world - 1)
+
     def _append_token(self, s: Sequence, tok: int) -> None:
         s.token_ids.append(tok)
         if s.sampling.stop_token is not None and tok == s.sampling.stop_token:
@@ -391,6 +403,7 @@ class Engine:
         hidden = self.model.forward_prefill(batch, self.kv_caches)
         logits = self.model.compute_logits(hidden)
         next_tokens = self._sample(logits, [s for s, _ in sel])
+        next_tokens = self._pp_sync_tokens(next_tokens)
         out = {}
         for i, (s, take) in enumerate(sel):
             s.num_cached += take
@@ -437,6 +450,7 @@ class Engine:
             logits = self.model.compute_logits(hidden)
             next_tokens = (logits.argmax(dim=-1) if all_greedy
                            else self._sample(logits, seqs))
+        next_tokens = self._pp_sync_tokens(next_tokens)
         out = {}
         for i, s in enumerate(seqs):
             tok = int(next_tokens[i])
