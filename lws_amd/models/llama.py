@@ -275,6 +275,12 @@ class LlamaForCausalLM:
         self.rope_table = build_rope_table(cfg.max_position, cfg.head_dim,
                                            cfg.rope_theta, device=self.device)
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        # TP collectives gate on the MODEL's tp_world, not the global
+        # process-group size: in PP mode the default group spans pipeline
+        # stages and a global all-reduce would mix stage activations
+        self._ar = all_reduce if tp_world > 1 else (lambda t: t)
+        self._agc = all_gather_cat if tp_world > 1 \
+            else (lambda t, dim=-1: t)
 
     # -- weights --------------------------------------------------------
     def materialize(self, seed: int = 0) -> int:
@@ -402,10 +408,10 @@ class LlamaForCausalLM:
             attn = self._prefill_attention(q, k, v, batch, layer,
                                            k_cache, v_cache)
             o = layer.o(attn)
-            x = all_reduce(o)
+            x = self._ar(o)
             # MLP
             h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)
-            x = all_reduce(self._mlp(layer, h))
+            x = self._ar(self._mlp(layer, h))
         B = len(batch.seq_starts) - 1
         if not self.pp_last:
             from ..parallel import pp as ppmod
@@ -496,9 +502,9 @@ class LlamaForCausalLM:
                     q.contiguous(), k_cache, v_cache, batch.block_tables,
                     batch.seq_lens, self.scale)
             o = layer.o(attn.view(B, layer.hq * cfg.head_dim))
-            x = all_reduce(o)
+            x = self._ar(o)
             h, residual = self._fused_add_rmsnorm(x, residual, layer.post_norm)
-            x = all_reduce(self._mlp(layer, h))
+            x = self._ar(self._mlp(layer, h))
         if not self.pp_last:
             from ..parallel import pp as ppmod
             ppmod.send_stage(torch.stack([x, residual]), self.pp_rank + 1)
@@ -564,4 +570,4 @@ class LlamaForCausalLM:
             return torch.zeros(hidden.size(0), self.cfg.vocab_size,
                                device=hidden.device, dtype=hidden.dtype)
         local = self.lm_head(hidden)
-        return all_gather_cat(local, dim=-1)
+        return self._agc(local, dim=-1)

@@ -29,6 +29,10 @@ def main(argv=None) -> int:
                    default=int(os.environ.get("LWS_AMD_HTTP_PORT", "8000")))
     p.add_argument("--device", default=None)
     p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--parallel", default=os.environ.get("LWS_AMD_PARALLEL",
+                                                        "tp"),
+                   choices=("tp", "pp"),
+                   help="shard the group by tensor or pipeline parallelism")
     args = p.parse_args(argv)
 
     import torch
@@ -67,7 +71,8 @@ def main(argv=None) -> int:
     conductor.command({"op": "build",
                        "spec": {"model": args.model,
                                 "kv_pages": args.kv_pages,
-                                "seed": args.seed}})
+                                "seed": args.seed,
+                                "parallel": args.parallel}})
     engine = conductor.host.engine
 
     from lws_amd.serving.runtime import CollectiveEngine

@@ -47,14 +47,18 @@ class ShardHost:
     def build(self, spec: dict) -> dict:
         self.teardown()
         import os
+        parallel = spec.get("parallel", "tp")   # "tp" | "pp" over the group
         cfg = EngineConfig(model=spec["model"],
                            kv_pages=int(spec.get("kv_pages", 128)),
                            seed=int(spec.get("seed", 0)),
                            weight_dtype=spec.get(
                                "weight_dtype",
                                os.environ.get("LWS_AMD_WEIGHT_DTYPE", "bf16")),
-                           device=self.device, tp_rank=self.rank,
-                           tp_world=self.world)
+                           device=self.device,
+                           tp_rank=self.rank if parallel == "tp" else 0,
+                           tp_world=self.world if parallel == "tp" else 1,
+                           pp_rank=self.rank if parallel == "pp" else 0,
+                           pp_world=self.world if parallel == "pp" else 1)
         self.engine = Engine(cfg)
         info = self.engine.load()
         self.timings = info

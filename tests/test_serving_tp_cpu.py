@@ -13,8 +13,7 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-@pytest.fixture
-def tp2_server():
+def _spawn_server(extra_args):
     http_port = random.randint(21000, 59000)
     master_port = random.randint(21000, 59000)
     proc = subprocess.Popen(
@@ -22,7 +21,7 @@ def tp2_server():
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
          "--master-port", str(master_port), "-m", "lws_amd.serving.launch",
          "--model", "llama-tiny", "--device", "cpu", "--kv-pages", "64",
-         "--port", str(http_port)],
+         "--port", str(http_port), *extra_args],
         cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
         text=True, start_new_session=True)
     import httpx
@@ -57,6 +56,16 @@ def tp2_server():
                 pass
 
 
+@pytest.fixture
+def tp2_server():
+    yield from _spawn_server([])
+
+
+@pytest.fixture
+def pp2_server():
+    yield from _spawn_server(["--parallel", "pp"])
+
+
 def test_tp2_http_serving(tp2_server):
     import httpx
 
@@ -82,3 +91,22 @@ def test_tp2_http_serving(tp2_server):
     assert len(r3.json()["choices"][0]["token_ids"]) == 4
     m = httpx.get(base + "/metrics", timeout=10)
     assert "lws_amd_engine_requests_total 3" in m.text
+
+
+def test_pp2_http_serving(pp2_server):
+    """Same frontend, pipeline-parallel group: leader (stage 0) serves
+    HTTP; stage 1 holds the tail layers + lm_head and its sampled tokens
+    broadcast back."""
+    import httpx
+
+    base = pp2_server
+    r = httpx.post(base + "/v1/completions",
+                   json={"prompt": [3, 1, 4, 1, 5], "max_tokens": 4},
+                   timeout=120)
+    assert r.status_code == 200, r.text
+    toks = r.json()["choices"][0]["token_ids"]
+    assert len(toks) == 4
+    r2 = httpx.post(base + "/v1/completions",
+                    json={"prompt": [3, 1, 4, 1, 5], "max_tokens": 4},
+                    timeout=120)
+    assert r2.json()["choices"][0]["token_ids"] == toks
