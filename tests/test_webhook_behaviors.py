@@ -227,3 +227,28 @@ def test_all_example_manifests_parse_and_validate():
                         [r.name for r in obj.spec.roles]
                     seen += 1
     assert seen >= 3, f"expected >=3 example CRs, saw {seen}"
+
+
+def test_subgroup_labels_leader_excluded(cluster):
+    """LeaderExcluded: the leader carries no subgroup labels; workers
+    1..size-1 partition into (size-1)/subGroupSize groups
+    (pod_webhook.go:249-255 LeaderExcluded arm)."""
+    from lws_amd.api.leaderworkerset import SubGroupPolicy
+
+    lws = make_lws(name="sgx", replicas=1, size=5)
+    lws.spec.leader_worker_template.sub_group_policy = SubGroupPolicy(
+        type="LeaderExcluded", sub_group_size=2)
+    cluster.store.create(lws)
+    _wait_available(cluster, "sgx")
+    pods = {p.metadata.name: p for p in cluster.store.list("Pod", "default")}
+    assert len(pods) == 5
+    assert lwsapi.SUBGROUP_INDEX_LABEL_KEY not in pods["sgx-0"].metadata.labels
+    # workers 1,2 -> subgroup 0; workers 3,4 -> subgroup 1
+    assert pods["sgx-0-1"].metadata.labels[
+        lwsapi.SUBGROUP_INDEX_LABEL_KEY] == "0"
+    assert pods["sgx-0-2"].metadata.labels[
+        lwsapi.SUBGROUP_INDEX_LABEL_KEY] == "0"
+    assert pods["sgx-0-3"].metadata.labels[
+        lwsapi.SUBGROUP_INDEX_LABEL_KEY] == "1"
+    assert pods["sgx-0-4"].metadata.labels[
+        lwsapi.SUBGROUP_INDEX_LABEL_KEY] == "1"
