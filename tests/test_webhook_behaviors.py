@@ -158,6 +158,27 @@ def test_recreate_group_after_start_gate():
         if still_pending:
             # the gate held: restart while pending must not recreate
             assert now == uids, "group must not recreate while pods pending"
+        # the restart simulation stomped readiness; a real kubelet would
+        # re-report the container ready once it comes back — FakeRuntime
+        # only marks ready once, so simulate that recovery here or the
+        # gated (not-recreated) pod stays unready forever
+        def recover(name):
+            from lws_amd.cluster.store import NotFoundError
+            for _ in range(50):
+                try:
+                    cur = c.store.get("Pod", "default", name)
+                except NotFoundError:
+                    return
+                if not cur.status.container_statuses:
+                    return
+                cur.status.phase = "Running"
+                cur.status.container_statuses[0].ready = True
+                try:
+                    c.store.update_status(cur)
+                    return
+                except ConflictError:
+                    time.sleep(0.01)
+        recover("ags-0-1")
         _wait_available(c, "ags", timeout=30)
         # re-snapshot from the settled state: the pre-Available snapshot
         # can catch a pod mid-recreate (1 or 3 entries) and then the
