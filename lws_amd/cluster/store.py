@@ -22,7 +22,6 @@ import time
 from typing import Any, Callable, Optional
 
 from ..api import serde
-from ..api.meta import ObjectMeta
 
 FOREGROUND_FINALIZER = "foregroundDeletion"
 

@@ -16,7 +16,6 @@ import math
 from dataclasses import dataclass
 import torch
 
-from ..parallel import tp as tpmod
 from ..parallel.tp import ShardedLinear, all_gather_cat, all_reduce
 
 

@@ -7,7 +7,6 @@ leader additionally runs the frontend.  The engine sizes its KV pool for
 """
 from __future__ import annotations
 
-import math
 import time
 from dataclasses import dataclass, field
 from typing import Optional
@@ -16,7 +15,6 @@ import torch
 
 from ..models.llama import (DecodeBatch, LlamaConfig, LlamaForCausalLM,
                             MODEL_PRESETS, PrefillBatch)
-from ..parallel import tp as tpmod
 
 PAGE_SIZE = 16
 

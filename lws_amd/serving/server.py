@@ -22,7 +22,7 @@ from concurrent.futures import Future
 from dataclasses import dataclass, field
 from typing import Optional
 
-from .engine import Engine, EngineConfig, SamplingParams
+from .engine import Engine, SamplingParams
 
 
 @dataclass
