@@ -374,3 +374,55 @@ def test_percent_zero_budget_rejected():
             partition=0, max_unavailable="10%", max_surge="0%"))
     with pytest.raises(InvalidError):
         validate_lws(lws, None)
+
+
+def test_rolling_update_partition_properties():
+    """Invariants over the partition calculus (hypothesis):
+    monotonic (never above the current partition), bounded in
+    [0, replicas], and all-ready+updated drives it to 0."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=300, deadline=None)
+    @given(st.data())
+    def run(data):
+        n = data.draw(st.integers(min_value=1, max_value=8))
+        states = [(data.draw(st.booleans()), data.draw(st.booleans()))
+                  for _ in range(n)]
+        replicas = data.draw(st.integers(min_value=1, max_value=n))
+        step = data.draw(st.integers(min_value=1, max_value=n))
+        cur = data.draw(st.integers(min_value=0, max_value=n))
+        out = rolling_update_partition(states, replicas, step, cur)
+        assert 0 <= out <= max(cur, 0)
+        assert out <= len(states)
+        if all(r and u for r, u in states) and cur >= 0:
+            # fully rolled: partition only walks down
+            assert out <= cur
+
+    run()
+
+
+def test_store_concurrent_update_linearizes():
+    """Optimistic concurrency: N threads x M conflict-retried increments
+    on one object must all land (no lost updates)."""
+    import threading
+
+    from lws_amd.cluster.store import Store
+    from tests.conftest import make_lws, retry_update
+
+    store = Store()
+    lws = make_lws(name="cc", replicas=0)
+    store.create(lws)
+    N, M = 8, 25
+
+    def worker():
+        for _ in range(M):
+            retry_update(store, "LeaderWorkerSet", "default", "cc",
+                         lambda o: setattr(o.spec, "replicas",
+                                           o.spec.replicas + 1))
+    ts = [threading.Thread(target=worker) for _ in range(N)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert store.get("LeaderWorkerSet", "default", "cc").spec.replicas == N * M
