@@ -165,6 +165,32 @@ def cmd_describe(args) -> int:
     return 0
 
 
+def cmd_wait(args) -> int:
+    """kubectl wait --for=condition=Available analogue (the DS e2e
+    waiters poll conditions the same way)."""
+    import time as _time
+    resource = _resource(args.resource)
+    cs = _client(args)
+    rc_name = {"leaderworkersets": "leader_worker_sets",
+               "disaggregatedsets": "disaggregated_sets"}[resource]
+    rc = getattr(cs, rc_name)(args.namespace)
+    want = args.for_condition.split("=")
+    cond_type = want[0]
+    cond_status = want[1] if len(want) > 1 else "True"
+    deadline = _time.time() + args.timeout
+    while _time.time() < deadline:
+        obj = rc.get(args.name)
+        if obj is not None:
+            for c in obj.status.conditions or []:
+                if c.type == cond_type and c.status == cond_status:
+                    print(f"{args.resource}/{args.name} condition met: "
+                          f"{cond_type}={cond_status}")
+                    return 0
+        _time.sleep(0.2)
+    sys.exit(f"error: timed out waiting for {cond_type}={cond_status} "
+             f"on {args.resource}/{args.name}")
+
+
 def cmd_scale(args) -> int:
     resource = _resource(args.resource)
     cs = _client(args)
@@ -204,6 +230,14 @@ def main(argv=None) -> int:
     sp.add_argument("resource")
     sp.add_argument("name")
     sp.set_defaults(fn=cmd_describe)
+
+    sp = sub.add_parser("wait")
+    sp.add_argument("resource")
+    sp.add_argument("name")
+    sp.add_argument("--for", dest="for_condition", default="Available",
+                    help="condition, e.g. Available or Available=True")
+    sp.add_argument("--timeout", type=float, default=120.0)
+    sp.set_defaults(fn=cmd_wait)
 
     sp = sub.add_parser("scale")
     sp.add_argument("resource")

@@ -135,6 +135,9 @@ def test_lwsctl_events_and_describe(manager_proc):
     d = ctl("describe", "lws", "desc-lws")
     assert "status:" in d and "Available" in d
 
+    w = ctl("wait", "lws", "desc-lws", "--for", "Available", "--timeout", "30")
+    assert "condition met" in w
+
 
 def test_informer(manager_proc):
     proc, base = manager_proc
