@@ -21,7 +21,24 @@ from typing import Optional
 from ..api.core import Pod
 from ..api.meta import selector_matches
 from .controller import Controller, Manager
-from .node import Node, node_gpu_capacity, pod_gpu_request
+from .node import Node, node_gpu_capacity, pod_gpu_request  # noqa: F401
+
+
+def _as_number(v):
+    try:
+        return float(v)
+    except (TypeError, ValueError):
+        return None
+
+
+def _pod_resource_requests(pod) -> dict:
+    out: dict = {}
+    for c in pod.spec.containers:
+        for res, val in (c.resources.requests or {}).items():
+            n = _as_number(val)
+            if n and n > 0:
+                out[res] = out.get(res, 0) + n
+    return out
 from .store import ConflictError, NotFoundError, Store
 
 POD_GROUP_ANNOTATION = "scheduling.k8s.io/group-name"
@@ -135,14 +152,22 @@ class Scheduler:
         for k, v in (pod.spec.node_selector or {}).items():
             if node.metadata.labels.get(k) != v:
                 return False
-        # GPU resource fit
-        req = pod_gpu_request(pod)
-        if req > 0:
-            used = sum(pod_gpu_request(p) for p in scheduled
-                       if p.node_name == node.metadata.name
-                       and p.metadata.deletion_timestamp is None)
-            if used + req > node_gpu_capacity(node):
-                return False
+        # resource fit: every countable resource the pod requests must
+        # fit the node's remaining capacity (kube-scheduler NodeResources
+        # analogue; amd.com/gpu is just the resource the examples use)
+        reqs = _pod_resource_requests(pod)
+        if reqs:
+            colocated = [p for p in scheduled
+                         if p.node_name == node.metadata.name
+                         and p.metadata.deletion_timestamp is None]
+            for res, req in reqs.items():
+                cap = _as_number((node.capacity or {}).get(res))
+                if cap is None:
+                    return False         # node doesn't offer this resource
+                used = sum(_pod_resource_requests(p).get(res, 0)
+                           for p in colocated)
+                if used + req > cap:
+                    return False
         # affinity
         aff = pod.spec.affinity
         if aff is not None:

@@ -101,3 +101,33 @@ def test_gang_binds_after_capacity_frees():
                  timeout=30)
     finally:
         c.stop()
+
+
+def test_generic_resource_fit():
+    """The scheduler fits ANY countable requested resource against node
+    capacity, not just amd.com/gpu (kube NodeResources analogue)."""
+    from lws_amd.cluster.cluster import LwsCluster, make_nodes
+    from tests.conftest import make_lws, wait_for
+
+    nodes = make_nodes(1, gpus_per_node=8)
+    nodes[0].capacity["example.com/nic"] = 2
+    c = LwsCluster(nodes=nodes).start()
+    try:
+        lws = make_lws(name="nic", replicas=1, size=3)
+        tmpl = lws.spec.leader_worker_template.worker_template
+        tmpl.spec.containers[0].resources.requests = {
+            "amd.com/gpu": 1, "example.com/nic": 1}
+        c.store.create(lws)
+
+        # only 2 NICs: at most 2 of the 3 pods can bind
+        def two_bound():
+            ps = c.store.list("Pod", "default")
+            bound = [p for p in ps if p.node_name]
+            return ps if len(ps) == 3 and len(bound) == 2 else None
+        wait_for(two_bound, desc="2 of 3 pods bound", timeout=20)
+        import time
+        time.sleep(0.3)
+        bound = [p for p in c.store.list("Pod", "default") if p.node_name]
+        assert len(bound) == 2      # third stays Pending on NIC exhaustion
+    finally:
+        c.stop()
