@@ -212,8 +212,20 @@ def test_ds_slices():
         for rs in cur.status.role_statuses:
             assert rs.replicas == 2
 
-        # slice scale-down removes slice-1 objects
+        # slice scale-UP adds slice-2 objects and re-reaches Available
         from tests.conftest import retry_update
+        retry_update(c.store, dsapi.KIND, "default", "my-ds",
+                     lambda o: setattr(o.spec, "slices", 3))
+
+        def three_slices():
+            lws_list = c.store.list(lwsapi.KIND, "default")
+            slices = {l.metadata.labels[dsapi.SLICE_LABEL_KEY]
+                      for l in lws_list}
+            return (lws_list if len(lws_list) == 6 and
+                    slices == {"0", "1", "2"} and ds_available(c) else None)
+        wait_for(three_slices, desc="slice scale-up to 3", timeout=90)
+
+        # slice scale-down removes slice-1/2 objects
         retry_update(c.store, dsapi.KIND, "default", "my-ds",
                      lambda o: setattr(o.spec, "slices", 1))
         def one_slice():
@@ -222,7 +234,7 @@ def test_ds_slices():
                 return None
             return all(l.metadata.labels[dsapi.SLICE_LABEL_KEY] == "0"
                        for l in lws_list) or None
-        wait_for(one_slice, desc="slice-1 cleanup", timeout=60)
+        wait_for(one_slice, desc="slice-1/2 cleanup", timeout=60)
     finally:
         c.stop()
 
