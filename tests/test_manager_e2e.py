@@ -183,6 +183,32 @@ def test_ds_over_http(manager_proc):
     lws_names = [o.metadata.name for o in
                  cs.leader_worker_sets("default").list()]
     assert len(lws_names) == 2
+    # HPA-on-DS path: an External role gets an auto-created RoleScaler
+    # whose /scale subresource drives the role's replicas over HTTP
+    from tests.conftest import retry_update as _ru  # noqa: F401
+    cur = ds_client.get("http-ds")
+    for role in cur.spec.roles:
+        if role.name == "decode":
+            from lws_amd.api.disaggregatedset import RoleScaling
+            role.scaling = RoleScaling(mode="External")
+    ds_client.update(cur)
+    scaler = cs.role_scalers("default")
+
+    def scaler_ready():
+        sc = scaler.get("http-ds-decode")
+        return sc if sc is not None else None
+    wait_for(scaler_ready, desc="auto-created RoleScaler", timeout=30)
+    scaler.scale("http-ds-decode", 2)
+
+    def decode_scaled():
+        cur2 = ds_client.get("http-ds")
+        rs = {r.name: r for r in cur2.status.role_statuses or []}
+        d = rs.get("decode")
+        return cur2 if d is not None and d.replicas == 2 else None
+    wait_for(decode_scaled, desc="decode scaled via /scale", timeout=60)
+    sc = scaler.get_scale("http-ds-decode")
+    assert sc["spec"]["replicas"] == 2
+
     ds_client.delete("http-ds")
     wait_for(lambda: not cs.leader_worker_sets("default").list(),
              desc="children GC'd", timeout=30)
