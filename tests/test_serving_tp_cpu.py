@@ -13,7 +13,7 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def _spawn_server(extra_args):
+def _spawn_server(extra_args, _attempt=0):
     http_port = random.randint(21000, 59000)
     master_port = random.randint(21000, 59000)
     proc = subprocess.Popen(
@@ -32,8 +32,12 @@ def _spawn_server(extra_args):
     try:
         while time.monotonic() < deadline:
             if proc.poll() is not None:
-                raise AssertionError(
-                    f"server died: {proc.stdout.read()[-3000:]}")
+                out = proc.stdout.read()
+                if "address already in use" in out and _attempt < 3:
+                    # random port collided with a concurrent test: retry
+                    yield from _spawn_server(extra_args, _attempt + 1)
+                    return
+                raise AssertionError(f"server died: {out[-3000:]}")
             try:
                 r = httpx.get(base + "/health", timeout=2)
                 if r.status_code == 200:
