@@ -13,25 +13,35 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 @pytest.fixture
 def manager_proc():
-    port = random.randint(20000, 60000)
-    proc = subprocess.Popen(
-        [sys.executable, "-m", "lws_amd", "--api-bind", f"127.0.0.1:{port}",
-         "--nodes", "1", "--scheduler-provider", "gang"],
-        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True)
-    base = f"http://127.0.0.1:{port}"
     from lws_amd.client.clientset import HttpTransport
 
-    t = HttpTransport(base)
-    deadline = time.monotonic() + 60
-    while time.monotonic() < deadline:
-        if proc.poll() is not None:
-            raise AssertionError(f"manager died: {proc.stderr.read()[-2000:]}")
-        if t.healthz():
+    proc = base = None
+    for attempt in range(4):
+        port = random.randint(20000, 60000)
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "lws_amd", "--api-bind",
+             f"127.0.0.1:{port}", "--nodes", "1",
+             "--scheduler-provider", "gang"],
+            cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True)
+        base = f"http://127.0.0.1:{port}"
+        t = HttpTransport(base)
+        deadline = time.monotonic() + 60
+        while time.monotonic() < deadline:
+            if proc.poll() is not None:
+                err = proc.stderr.read()
+                if "address already in use" in err and attempt < 3:
+                    proc = None
+                    break           # random port collided: retry
+                raise AssertionError(f"manager died: {err[-2000:]}")
+            if t.healthz():
+                break
+            time.sleep(0.1)
+        else:
+            proc.kill()
+            raise AssertionError("manager never became healthy")
+        if proc is not None:
             break
-        time.sleep(0.1)
-    else:
-        proc.kill()
-        raise AssertionError("manager never became healthy")
     yield proc, base
     proc.terminate()
     try:
