@@ -208,7 +208,19 @@ def test_ds_over_http(manager_proc):
         sc = scaler.get("http-ds-decode")
         return sc if sc is not None else None
     wait_for(scaler_ready, desc="auto-created RoleScaler", timeout=30)
-    scaler.scale("http-ds-decode", 2)
+
+    def do_scale():
+        # the /scale PUT races the scaler-manager's status writes: retry
+        # on conflict like any HPA client would
+        import httpx
+        try:
+            scaler.scale("http-ds-decode", 2)
+            return True
+        except httpx.HTTPStatusError as e:
+            if e.response.status_code == 409:
+                return None
+            raise
+    wait_for(do_scale, desc="scale accepted", timeout=30)
 
     def decode_scaled():
         cur2 = ds_client.get("http-ds")
