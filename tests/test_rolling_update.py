@@ -250,7 +250,23 @@ def test_restart_policy_recreates_group(cluster):
                     "restart-0-2"):
                 agent.mark_container_restarted(w)
         return None
-    wait_for(recreated, desc="group recreated with new pods", timeout=30)
+    try:
+        wait_for(recreated, desc="group recreated with new pods", timeout=30)
+    except AssertionError:
+        for pod in cluster.store.list("Pod", "default"):
+            cs = pod.status.container_statuses
+            print(f"DUMP pod={pod.metadata.name} uid={pod.metadata.uid} "
+                  f"old_uid={uids_before.get(pod.metadata.name)} "
+                  f"del_ts={pod.metadata.deletion_timestamp} "
+                  f"fin={pod.metadata.finalizers} phase={pod.status.phase} "
+                  f"restarts={[c.restart_count for c in cs or []]}")
+        for sts in cluster.store.list("StatefulSet", "default"):
+            print(f"DUMP sts={sts.metadata.name} "
+                  f"del_ts={sts.metadata.deletion_timestamp} "
+                  f"replicas={sts.spec.replicas}")
+        for ev in cluster.store.list("Event", "default"):
+            print(f"DUMP event {ev.reason} x{ev.count}: {ev.message}")
+        raise
     _wait_available(cluster, "restart")
 
 
