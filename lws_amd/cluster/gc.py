@@ -31,20 +31,29 @@ class GarbageCollector:
             self.ctrl.enqueue("", "sweep")
 
     def reconcile(self, namespace: str, name: str) -> Optional[float]:
-        live_uids = set()
+        live = {}
         objects = []
         for kind, ns, obj_name in self.store.snapshot_keys():
             obj = self.store.try_get(kind, ns, obj_name)
             if obj is None:
                 continue
-            live_uids.add(obj.metadata.uid)
+            live[obj.metadata.uid] = obj
             objects.append(obj)
         for obj in objects:
             ctrl_ref = next((r for r in obj.metadata.owner_references
                              if r.controller), None)
-            if ctrl_ref is None or ctrl_ref.uid in live_uids:
+            if ctrl_ref is None:
                 continue
             if obj.metadata.deletion_timestamp is not None:
+                continue
+            owner = live.get(ctrl_ref.uid)
+            # collect when the controller owner is gone, OR is itself being
+            # deleted (kube foregroundDeletion GC semantics): a dependent
+            # created in the TOCTOU window after the cascade pass — e.g. a
+            # StatefulSet reconcile re-creating a pod for a just-marked-
+            # deleting STS — would otherwise block the foreground chain
+            # forever
+            if owner is not None and owner.metadata.deletion_timestamp is None:
                 continue
             try:
                 self.store.delete(type(obj).__name__
